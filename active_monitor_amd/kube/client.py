@@ -1,0 +1,139 @@
+"""Async Kubernetes client facade.
+
+The reconciler programs against this interface the way the reference programs
+against controller-runtime's ``client.Client`` + the dynamic client + the typed
+clientset (healthcheck_controller.go:133-137). Backends:
+
+- :class:`MemoryClient` — wraps :class:`~active_monitor_amd.kube.memory.MemoryApiServer`
+  (the envtest/bench backend), with optional simulated per-op latency so
+  benchmarks can model apiserver RTT honestly.
+- ``HttpClient`` (kube/http.py) — a real apiserver over HTTP(S).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Any, Dict, List, Optional, Protocol
+
+from .memory import MemoryApiServer, Subscription
+
+Obj = Dict[str, Any]
+
+
+class KubeClient(Protocol):
+    async def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj: ...
+
+    async def list(
+        self, api_version: str, kind: str,
+        namespace: Optional[str] = None, label_selector: Optional[str] = None,
+    ) -> List[Obj]: ...
+
+    async def create(self, obj: Obj) -> Obj: ...
+
+    async def update(self, obj: Obj) -> Obj: ...
+
+    async def update_status(self, obj: Obj) -> Obj: ...
+
+    async def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None: ...
+
+    def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription: ...
+
+
+class MemoryClient:
+    """KubeClient over the in-memory apiserver.
+
+    ``latency`` (seconds) is awaited before every request to emulate a real
+    apiserver round-trip; 0 disables it.
+    """
+
+    def __init__(self, server: Optional[MemoryApiServer] = None, latency: float = 0.0):
+        self.server = server or MemoryApiServer()
+        self.latency = latency
+
+    async def _lat(self) -> None:
+        if self.latency > 0:
+            await asyncio.sleep(self.latency)
+        else:
+            # cooperative yield so tight loops don't starve the event loop
+            await asyncio.sleep(0)
+
+    async def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+        await self._lat()
+        return self.server.get(api_version, kind, namespace, name)
+
+    async def list(
+        self, api_version: str, kind: str,
+        namespace: Optional[str] = None, label_selector: Optional[str] = None,
+    ) -> List[Obj]:
+        await self._lat()
+        return self.server.list(api_version, kind, namespace, label_selector)
+
+    async def create(self, obj: Obj) -> Obj:
+        await self._lat()
+        return self.server.create(obj)
+
+    async def update(self, obj: Obj) -> Obj:
+        await self._lat()
+        return self.server.update(obj)
+
+    async def update_status(self, obj: Obj) -> Obj:
+        await self._lat()
+        return self.server.update_status(obj)
+
+    async def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None:
+        await self._lat()
+        self.server.delete(api_version, kind, namespace, name)
+
+    def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription:
+        return self.server.watch(api_version, kind, namespace)
+
+
+class EventRecorder:
+    """Kubernetes Event emission (record.EventRecorder equivalent — the
+    reference emits events on nearly every transition, e.g.
+    healthcheck_controller.go:243,280,532,636,663)."""
+
+    NORMAL = "Normal"
+    WARNING = "Warning"
+
+    def __init__(self, client: KubeClient, component: str = "active-monitor"):
+        self.client = client
+        self.component = component
+
+    async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
+        meta = involved.get("metadata") or {}
+        ns = meta.get("namespace", "") or "default"
+        ev = {
+            "apiVersion": "v1",
+            "kind": "Event",
+            "metadata": {"generateName": (meta.get("name", "object") + "."), "namespace": ns},
+            "involvedObject": {
+                "apiVersion": involved.get("apiVersion"),
+                "kind": involved.get("kind"),
+                "name": meta.get("name"),
+                "namespace": meta.get("namespace", ""),
+                "uid": meta.get("uid", ""),
+            },
+            "reason": reason,
+            "message": message,
+            "type": ev_type,
+            "source": {"component": self.component},
+            "firstTimestamp": None,
+        }
+        try:
+            await self.client.create(ev)
+        except Exception:  # events are best-effort, never fail the caller
+            pass
+
+
+class FakeRecorder(EventRecorder):
+    """Test recorder capturing events in-memory
+    (record.NewFakeRecorder equivalent, used across the reference's unit
+    tests, healthcheck_controller_unit_test.go:40-46)."""
+
+    def __init__(self, capacity: int = 100):
+        self.events: List[str] = []
+        self.capacity = capacity
+
+    async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
+        if len(self.events) < self.capacity:
+            self.events.append(f"{ev_type} {reason} {message}")

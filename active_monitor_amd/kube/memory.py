@@ -1,0 +1,334 @@
+"""In-memory Kubernetes apiserver — the framework's envtest equivalent.
+
+The reference's integration tier runs against envtest (a real kube-apiserver +
+etcd, suite_test.go:67-134). This module provides the same role as a
+dependency-free in-process store with the apiserver semantics the controller
+exercises:
+
+- optimistic concurrency via ``metadata.resourceVersion`` (conflict errors on
+  stale writes, so the reconciler's retry-on-conflict paths are real),
+- ``metadata.generateName`` server-side name generation,
+- the HealthCheck status subresource (plain updates cannot touch status and
+  vice versa),
+- ownerReference cascade GC (Workflows vanish when their HealthCheck is
+  deleted — the behavior relied on at healthcheck_controller.go:512-522),
+- finalizer-driven deletionTimestamp handling,
+- label-selector list filtering and watch streams (ADDED/MODIFIED/DELETED).
+
+The core is synchronous and thread-safe; watches are asyncio queues fed via
+``call_soon_threadsafe`` so any thread may mutate the store.
+"""
+from __future__ import annotations
+
+import asyncio
+import copy
+import random
+import string
+import threading
+from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
+
+from ..api.types import k8s_now
+from .errors import AlreadyExistsError, ConflictError, InvalidError, NotFoundError
+from .registry import DEFAULT_REGISTRY, STATUS_SUBRESOURCE_KINDS, Registry
+
+Obj = Dict[str, Any]
+Key = Tuple[str, str, str, str]  # (apiVersion, kind, namespace, name)
+
+_SUFFIX_ALPHABET = "bcdfghjklmnpqrstvwxz2456789"  # k8s-style name suffix chars
+
+
+def _rand_suffix(n: int = 5) -> str:
+    return "".join(random.choice(_SUFFIX_ALPHABET) for _ in range(n))
+
+
+def parse_label_selector(selector: Optional[str]) -> Dict[str, str]:
+    """Parse an equality-based selector string ``k=v,k2=v2``."""
+    if not selector:
+        return {}
+    out: Dict[str, str] = {}
+    for part in selector.split(","):
+        part = part.strip()
+        if not part:
+            continue
+        if "=" not in part:
+            raise InvalidError(f"unsupported selector term: {part!r}")
+        k, _, v = part.partition("=")
+        out[k.strip()] = v.lstrip("=").strip()  # tolerate '=='
+    return out
+
+
+def _labels_match(obj: Obj, selector: Dict[str, str]) -> bool:
+    labels = (obj.get("metadata") or {}).get("labels") or {}
+    return all(labels.get(k) == v for k, v in selector.items())
+
+
+class Subscription:
+    """One watch stream. Async-iterate to receive ``{"type": ..., "object": ...}``."""
+
+    def __init__(self, server: "MemoryApiServer", api_version: str, kind: str,
+                 namespace: Optional[str], loop: asyncio.AbstractEventLoop):
+        self._server = server
+        self.api_version = api_version
+        self.kind = kind
+        self.namespace = namespace
+        self._loop = loop
+        self._queue: "asyncio.Queue[Optional[Dict[str, Any]]]" = asyncio.Queue()
+        self._closed = False
+
+    def _offer(self, event: Dict[str, Any]) -> None:
+        if self._closed:
+            return
+        obj = event["object"]
+        meta = obj.get("metadata") or {}
+        if obj.get("apiVersion") != self.api_version or obj.get("kind") != self.kind:
+            return
+        if self.namespace is not None and meta.get("namespace", "") != self.namespace:
+            return
+        try:
+            self._loop.call_soon_threadsafe(self._queue.put_nowait, copy.deepcopy(event))
+        except RuntimeError:
+            self._closed = True  # loop gone
+
+    def close(self) -> None:
+        if not self._closed:
+            self._closed = True
+            self._server._unsubscribe(self)
+            try:
+                self._loop.call_soon_threadsafe(self._queue.put_nowait, None)
+            except RuntimeError:
+                pass
+
+    def __aiter__(self) -> AsyncIterator[Dict[str, Any]]:
+        return self
+
+    async def __anext__(self) -> Dict[str, Any]:
+        ev = await self._queue.get()
+        if ev is None:
+            raise StopAsyncIteration
+        return ev
+
+
+class MemoryApiServer:
+    def __init__(self, registry: Registry = DEFAULT_REGISTRY):
+        self.registry = registry
+        self._objects: Dict[Key, Obj] = {}
+        self._rv = 0
+        self._lock = threading.RLock()
+        self._subs: List[Subscription] = []
+        # counters for observability/benchmarks
+        self.op_counts: Dict[str, int] = {"get": 0, "list": 0, "create": 0,
+                                          "update": 0, "update_status": 0, "delete": 0}
+
+    # -- internals ---------------------------------------------------------
+
+    def _next_rv(self) -> str:
+        self._rv += 1
+        return str(self._rv)
+
+    def _key(self, obj: Obj) -> Key:
+        meta = obj.get("metadata") or {}
+        info = self.registry.by_kind(obj.get("apiVersion", ""), obj.get("kind", ""))
+        ns = meta.get("namespace", "") if info.namespaced else ""
+        return (obj.get("apiVersion", ""), obj.get("kind", ""), ns, meta.get("name", ""))
+
+    def _not_found(self, api_version: str, kind: str, name: str) -> NotFoundError:
+        info = self.registry.by_kind(api_version, kind)
+        group = api_version.split("/")[0] if "/" in api_version else ""
+        full = f"{info.plural}.{group}" if group else info.plural
+        return NotFoundError(f'{full} "{name}" not found')
+
+    def _publish(self, ev_type: str, obj: Obj) -> None:
+        event = {"type": ev_type, "object": obj}
+        for sub in list(self._subs):
+            sub._offer(event)
+
+    def _unsubscribe(self, sub: Subscription) -> None:
+        with self._lock:
+            if sub in self._subs:
+                self._subs.remove(sub)
+
+    # -- public API --------------------------------------------------------
+
+    def create(self, obj: Obj) -> Obj:
+        obj = copy.deepcopy(obj)
+        meta = obj.setdefault("metadata", {})
+        with self._lock:
+            self.op_counts["create"] += 1
+            if not meta.get("name"):
+                gen = meta.get("generateName")
+                if not gen:
+                    raise InvalidError("name or generateName is required")
+                # retry suffixes on collision, like the apiserver
+                for _ in range(16):
+                    candidate = gen + _rand_suffix()
+                    meta["name"] = candidate
+                    if self._key(obj) not in self._objects:
+                        break
+                else:
+                    raise AlreadyExistsError(f"could not generate unique name for {gen}")
+            key = self._key(obj)
+            if key in self._objects:
+                raise AlreadyExistsError(
+                    f'{obj.get("kind", "object")} "{meta["name"]}" already exists'
+                )
+            meta["uid"] = meta.get("uid") or ("uid-" + _rand_suffix(12))
+            meta["resourceVersion"] = self._next_rv()
+            meta["creationTimestamp"] = meta.get("creationTimestamp") or k8s_now()
+            meta["generation"] = 1
+            self._objects[key] = obj
+            out = copy.deepcopy(obj)
+            self._publish("ADDED", out)
+        return out
+
+    def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+        info = self.registry.by_kind(api_version, kind)
+        ns = namespace if info.namespaced else ""
+        with self._lock:
+            self.op_counts["get"] += 1
+            obj = self._objects.get((api_version, kind, ns, name))
+            if obj is None:
+                raise self._not_found(api_version, kind, name)
+            return copy.deepcopy(obj)
+
+    def list(
+        self,
+        api_version: str,
+        kind: str,
+        namespace: Optional[str] = None,
+        label_selector: Optional[str] = None,
+    ) -> List[Obj]:
+        selector = parse_label_selector(label_selector)
+        with self._lock:
+            self.op_counts["list"] += 1
+            out = []
+            for (av, k, ns, _), obj in self._objects.items():
+                if av != api_version or k != kind:
+                    continue
+                if namespace is not None and ns != namespace:
+                    continue
+                if selector and not _labels_match(obj, selector):
+                    continue
+                out.append(copy.deepcopy(obj))
+            return out
+
+    def update(self, obj: Obj) -> Obj:
+        obj = copy.deepcopy(obj)
+        key = self._key(obj)
+        meta = obj.setdefault("metadata", {})
+        with self._lock:
+            self.op_counts["update"] += 1
+            existing = self._objects.get(key)
+            if existing is None:
+                raise self._not_found(key[0], key[1], key[3])
+            ex_meta = existing["metadata"]
+            rv = meta.get("resourceVersion")
+            if rv and str(rv) != str(ex_meta.get("resourceVersion")):
+                raise ConflictError(
+                    f'Operation cannot be fulfilled on {obj.get("kind")} '
+                    f'"{meta.get("name")}": the object has been modified; please apply '
+                    f"your changes to the latest version and try again"
+                )
+            # immutable metadata
+            meta["uid"] = ex_meta.get("uid")
+            meta["creationTimestamp"] = ex_meta.get("creationTimestamp")
+            if ex_meta.get("deletionTimestamp"):
+                meta["deletionTimestamp"] = ex_meta["deletionTimestamp"]
+            # status subresource: plain update cannot change status
+            if (key[0], key[1]) in STATUS_SUBRESOURCE_KINDS:
+                if "status" in existing:
+                    obj["status"] = copy.deepcopy(existing["status"])
+                else:
+                    obj.pop("status", None)
+            if obj.get("spec") != existing.get("spec"):
+                meta["generation"] = int(ex_meta.get("generation", 1)) + 1
+            else:
+                meta["generation"] = ex_meta.get("generation", 1)
+            meta["resourceVersion"] = self._next_rv()
+            self._objects[key] = obj
+            # finalizer removal completes a pending delete
+            if meta.get("deletionTimestamp") and not meta.get("finalizers"):
+                del self._objects[key]
+                out = copy.deepcopy(obj)
+                self._publish("DELETED", out)
+                return out
+            out = copy.deepcopy(obj)
+            self._publish("MODIFIED", out)
+        return out
+
+    def update_status(self, obj: Obj) -> Obj:
+        obj = copy.deepcopy(obj)
+        key = self._key(obj)
+        meta = obj.get("metadata") or {}
+        with self._lock:
+            self.op_counts["update_status"] += 1
+            existing = self._objects.get(key)
+            if existing is None:
+                raise self._not_found(key[0], key[1], key[3])
+            ex_meta = existing["metadata"]
+            rv = meta.get("resourceVersion")
+            if rv and str(rv) != str(ex_meta.get("resourceVersion")):
+                raise ConflictError(
+                    f'Operation cannot be fulfilled on {obj.get("kind")} '
+                    f'"{meta.get("name")}": the object has been modified; please apply '
+                    f"your changes to the latest version and try again"
+                )
+            updated = copy.deepcopy(existing)
+            if "status" in obj:
+                updated["status"] = copy.deepcopy(obj["status"])
+            else:
+                updated.pop("status", None)
+            updated["metadata"]["resourceVersion"] = self._next_rv()
+            self._objects[key] = updated
+            out = copy.deepcopy(updated)
+            self._publish("MODIFIED", out)
+        return out
+
+    def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None:
+        info = self.registry.by_kind(api_version, kind)
+        ns = namespace if info.namespaced else ""
+        with self._lock:
+            self.op_counts["delete"] += 1
+            key = (api_version, kind, ns, name)
+            obj = self._objects.get(key)
+            if obj is None:
+                raise self._not_found(api_version, kind, name)
+            meta = obj["metadata"]
+            if meta.get("finalizers"):
+                if not meta.get("deletionTimestamp"):
+                    meta["deletionTimestamp"] = k8s_now()
+                    meta["resourceVersion"] = self._next_rv()
+                    self._publish("MODIFIED", copy.deepcopy(obj))
+                return
+            del self._objects[key]
+            self._publish("DELETED", copy.deepcopy(obj))
+            self._cascade_delete(meta.get("uid"))
+
+    def _cascade_delete(self, owner_uid: Optional[str]) -> None:
+        """Background-propagation GC: delete dependents whose ownerReferences
+        name the deleted uid (the mechanism behind Workflow cleanup on
+        HealthCheck delete, healthcheck_controller.go:512-522)."""
+        if not owner_uid:
+            return
+        dependents = [
+            key for key, o in self._objects.items()
+            if any(
+                ref.get("uid") == owner_uid
+                for ref in (o.get("metadata") or {}).get("ownerReferences") or []
+            )
+        ]
+        for key in dependents:
+            obj = self._objects.pop(key, None)
+            if obj is not None:
+                self._publish("DELETED", copy.deepcopy(obj))
+                self._cascade_delete(obj["metadata"].get("uid"))
+
+    def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription:
+        loop = asyncio.get_running_loop()
+        sub = Subscription(self, api_version, kind, namespace, loop)
+        with self._lock:
+            self._subs.append(sub)
+        return sub
+
+    def __len__(self) -> int:
+        with self._lock:
+            return len(self._objects)
