@@ -1,0 +1,165 @@
+"""Controller manager — the controller-runtime equivalent.
+
+Wires what the reference's ``cmd/main.go`` + controller-runtime provide:
+
+- an informer: initial list + watch of HealthCheck CRs feeding the workqueue
+  (the apiserver watch streams controller-runtime maintains),
+- ``max_workers`` reconcile workers (``MaxConcurrentReconciles``,
+  healthcheck_controller.go:298, flag ``max-workers`` default 10,
+  cmd/main.go:144) — THE 1/2/4/8 scaling knob of BASELINE.md,
+- error policy: reconcile errors are re-queued with per-item exponential
+  backoff; ``requeue_after`` honors the reference's 1s error requeue (:204),
+- /healthz and /readyz probes plus the Prometheus /metrics endpoint
+  (cmd/main.go:74-85,121-126),
+- optional Lease-based leader election (cmd/main.go:87-88).
+
+Per-reconcile latencies are recorded (monotonic clock) for the benchmark
+harness — this is the instrumentation BASELINE.md's p50 metric reads.
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from typing import List, Optional, Tuple
+
+from .. import API_VERSION
+from ..kube.client import EventRecorder, KubeClient
+from .leader import LeaderElector
+from .reconciler import HC_KIND, HealthCheckReconciler
+from .workqueue import WorkQueue
+
+log = logging.getLogger("active_monitor_amd.manager")
+
+
+class Manager:
+    def __init__(
+        self,
+        client: KubeClient,
+        max_workers: int = 10,
+        recorder: Optional[EventRecorder] = None,
+        namespace: Optional[str] = None,
+        metrics_addr: Optional[Tuple[str, int]] = None,
+        health_addr: Optional[Tuple[str, int]] = None,
+        leader_elect: bool = False,
+        leader_identity: str = "",
+        record_latencies: bool = True,
+    ):
+        self.client = client
+        self.max_workers = max_workers
+        self.namespace = namespace
+        self.queue = WorkQueue()
+        self.recorder = recorder or EventRecorder(client)
+        self.reconciler = HealthCheckReconciler(
+            client, self.recorder, max_parallel=max_workers, queue=self.queue
+        )
+        self.metrics_addr = metrics_addr
+        self.health_addr = health_addr
+        self.leader_elect = leader_elect
+        self.leader_identity = leader_identity
+        self._tasks: List[asyncio.Task] = []
+        self._sub = None
+        self._started = asyncio.Event()
+        self._stopped = False
+        self._servers: List[object] = []
+        self.record_latencies = record_latencies
+        self.latencies: List[float] = []
+
+    # -- lifecycle ----------------------------------------------------------
+
+    async def start(self) -> None:
+        """Start informer + workers (+ endpoints); returns once running."""
+        if self.leader_elect:
+            elector = LeaderElector(
+                self.client,
+                # LeaderElectionID from the reference (cmd/main.go:88)
+                name="689451f8.keikoproj.io",
+                namespace=self.namespace or "default",
+                identity=self.leader_identity or f"manager-{id(self):x}",
+            )
+            await elector.acquire()
+            self._tasks.append(asyncio.ensure_future(elector.renew_loop()))
+
+        if self.health_addr is not None or self.metrics_addr is not None:
+            from .endpoints import serve_endpoints
+
+            self._servers = await serve_endpoints(
+                self, health=self.health_addr, metrics=self.metrics_addr
+            )
+
+        self._tasks.append(asyncio.ensure_future(self._informer()))
+        for i in range(self.max_workers):
+            self._tasks.append(asyncio.ensure_future(self._worker(i)))
+        self._started.set()
+
+    async def run_forever(self) -> None:
+        await self.start()
+        try:
+            await asyncio.gather(*self._tasks)
+        except asyncio.CancelledError:
+            pass
+
+    async def stop(self) -> None:
+        self._stopped = True
+        if self._sub is not None:
+            self._sub.close()
+        await self.queue.shutdown()
+        self.reconciler.stop_all()
+        for t in self._tasks:
+            t.cancel()
+        await asyncio.gather(*self._tasks, return_exceptions=True)
+        for srv in self._servers:
+            srv.close()
+
+    @property
+    def ready(self) -> bool:
+        return self._started.is_set() and not self._stopped
+
+    # -- informer -----------------------------------------------------------
+
+    async def _informer(self) -> None:
+        self._sub = self.client.watch(API_VERSION, HC_KIND, self.namespace)
+        # initial list AFTER subscribing so no event can slip between the two
+        for obj in await self.client.list(API_VERSION, HC_KIND, self.namespace):
+            meta = obj.get("metadata") or {}
+            await self.queue.add((meta.get("namespace", ""), meta.get("name", "")))
+        async for ev in self._sub:
+            meta = ev["object"].get("metadata") or {}
+            await self.queue.add((meta.get("namespace", ""), meta.get("name", "")))
+
+    # -- workers ------------------------------------------------------------
+
+    async def _worker(self, idx: int) -> None:
+        while True:
+            item = await self.queue.get()
+            if item is None:
+                return
+            key, flags = item
+            ns, name = key
+            t0 = time.monotonic()
+            result = None
+            try:
+                result = await self.reconciler.reconcile(ns, name, flags)
+            except asyncio.CancelledError:
+                await self.queue.done(key)
+                raise
+            except Exception as e:  # reconciler already guards; belt & braces
+                log.error("worker %d: reconcile %s/%s raised: %s", idx, ns, name, e)
+            finally:
+                await self.queue.done(key)
+            if self.record_latencies:
+                self.latencies.append(time.monotonic() - t0)
+            if result is not None and result.error is not None:
+                # controller-runtime: error ⇒ rate-limited requeue
+                await self.queue.add_rate_limited(key)
+            elif result is not None and result.requeue_after > 0:
+                await self.queue.add_after(key, result.requeue_after)
+            else:
+                self.queue.forget(key)
+
+    # -- bench/test helpers --------------------------------------------------
+
+    def drain_latencies(self) -> List[float]:
+        out = self.latencies
+        self.latencies = []
+        return out

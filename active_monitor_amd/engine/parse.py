@@ -1,0 +1,138 @@
+"""Workflow YAML parsing + field injection.
+
+Re-implements the reference's parseWorkflowFromHealthcheck /
+parseRemedyWorkflowFromHealthcheck (healthcheck_controller.go:876-1125):
+
+- read the definition through the artifact store,
+- YAML-decode; non-map documents and missing/non-map ``spec`` are errors with
+  the reference's messages,
+- label handling: workflow-defined ``metadata.labels`` are used (values
+  stringified), the Argo instance-id label
+  ``workflows.argoproj.io/controller-instanceid: activemonitor-workflows``
+  is guaranteed as a default. Unlike the reference — whose shared
+  ``workflowLabels`` map leaks labels across HealthChecks —
+  labels here are scoped per submission,
+- inject ``spec.podGC = {strategy: OnPodCompletion}`` when unset (:953-977),
+- inject ``spec.serviceAccountName`` from the resource (:988-991),
+- ``spec.activeDeadlineSeconds`` defaulting: health-check workflows default to
+  ``Workflow.Timeout``, itself defaulted to ``RepeatAfterSec`` (a spec mutation
+  that later feeds the backoff computation, :980-995); remedy workflows default
+  to ``RepeatAfterSec`` and round-trip an existing numeric deadline back into
+  ``RemedyWorkflow.Timeout`` (:1108-1120).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, Tuple
+
+import yaml
+
+from ..api.types import HealthCheck
+from ..store import get_artifact_reader
+
+WF_INSTANCE_ID_LABEL_KEY = "workflows.argoproj.io/controller-instanceid"
+WF_INSTANCE_ID = "activemonitor-workflows"
+POD_GC_ON_POD_COMPLETION = "OnPodCompletion"
+
+
+class WorkflowParseError(Exception):
+    pass
+
+
+def _extract_labels(data: Dict[str, Any]) -> Dict[str, str]:
+    """Per-submission label derivation with the instance-id default."""
+    labels: Dict[str, str] = {}
+    metadata = data.get("metadata")
+    if isinstance(metadata, dict):
+        raw = metadata.get("labels")
+        if isinstance(raw, dict):
+            for k, v in raw.items():
+                labels[str(k)] = str(v)
+    labels.setdefault(WF_INSTANCE_ID_LABEL_KEY, WF_INSTANCE_ID)
+    return labels
+
+
+def _decode(content: bytes, remedy: bool) -> Dict[str, Any]:
+    try:
+        data = yaml.safe_load(content)
+    except yaml.YAMLError as e:
+        raise WorkflowParseError(f"Invalid spec file passed: {e}") from e
+    if data is None:
+        data = {}
+    if not isinstance(data, dict):
+        raise WorkflowParseError("Invalid spec file passed: not a mapping")
+    return data
+
+
+def _validated_spec(data: Dict[str, Any], remedy: bool) -> Dict[str, Any]:
+    spec_raw = data.get("spec")
+    if spec_raw is None:
+        raise WorkflowParseError(
+            "Invalid remedy workflow, missing spec" if remedy else "invalid workflow, missing spec"
+        )
+    if not isinstance(spec_raw, dict):
+        raise WorkflowParseError(
+            "invalid remedy workflow, spec is not a map"
+            if remedy
+            else "invalid workflow, spec is not a map"
+        )
+    return spec_raw
+
+
+def parse_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, Any], Dict[str, str]]:
+    """Returns ``(spec_dict, labels)`` for the health-check workflow, mutating
+    ``hc.spec.workflow.timeout`` when it defaults from RepeatAfterSec
+    (reference :980-995)."""
+    content = b""
+    if hc.spec.workflow.resource is not None:
+        reader = get_artifact_reader(hc.spec.workflow.resource.source)
+        content = reader.read()
+    data = _decode(content, remedy=False)
+    labels = _extract_labels(data)
+    spec = _validated_spec(data, remedy=False)
+
+    if spec.get("podGC") is None:
+        spec["podGC"] = {"strategy": POD_GC_ON_POD_COMPLETION}
+
+    if hc.spec.workflow.timeout == 0:
+        hc.spec.workflow.timeout = hc.spec.repeat_after_sec
+    timeout = hc.spec.workflow.timeout
+
+    if hc.spec.workflow.resource is not None and hc.spec.workflow.resource.service_account:
+        spec["serviceAccountName"] = hc.spec.workflow.resource.service_account
+
+    if spec.get("activeDeadlineSeconds") is None:
+        spec["activeDeadlineSeconds"] = timeout
+    return spec, labels
+
+
+def parse_remedy_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, Any], Dict[str, str]]:
+    """Remedy variant (reference :1002-1125): ``activeDeadlineSeconds``
+    defaults from RepeatAfterSec; an existing numeric deadline round-trips
+    into ``hc.spec.remedy_workflow.timeout``."""
+    content = b""
+    if hc.spec.remedy_workflow.resource is not None:
+        reader = get_artifact_reader(hc.spec.remedy_workflow.resource.source)
+        content = reader.read()
+    data = _decode(content, remedy=True)
+    labels = _extract_labels(data)
+    spec = _validated_spec(data, remedy=True)
+
+    if spec.get("podGC") is None:
+        spec["podGC"] = {"strategy": POD_GC_ON_POD_COMPLETION}
+
+    if (
+        hc.spec.remedy_workflow.resource is not None
+        and hc.spec.remedy_workflow.resource.service_account
+    ):
+        spec["serviceAccountName"] = hc.spec.remedy_workflow.resource.service_account
+
+    timeout = hc.spec.repeat_after_sec
+    deadline = spec.get("activeDeadlineSeconds")
+    if deadline is None:
+        spec["activeDeadlineSeconds"] = timeout
+        hc.spec.remedy_workflow.timeout = timeout
+    elif isinstance(deadline, (int, float)) and not isinstance(deadline, bool):
+        hc.spec.remedy_workflow.timeout = int(deadline)
+    else:
+        hc.spec.remedy_workflow.timeout = timeout
+    return spec, labels

@@ -1,0 +1,309 @@
+"""RBAC provisioning for health-check and remedy workflows.
+
+Re-implements the reference's SA/Role/ClusterRole/Binding lifecycle
+(healthcheck_controller.go:302-474, 1127-1443):
+
+- default least-privilege rule sets: read-only for health checks, scoped CRUD
+  for remedies (:85-120), overridable per-CR via ``spec.*.rbacRules`` (:124-129),
+- derived names ``<sa>-cluster-role``, ``<sa>-cluster-role-binding``,
+  ``<sa>-ns-role``, ``<sa>-ns-role-binding`` (:306-309,321-324),
+- get-then-create idempotency (existing objects are reused, never updated),
+- every created object labeled ``workflows.argoproj.io/managed-by:
+  active-monitor``; deletes only touch objects carrying that label
+  (:1169,1242,1317,1375,1433),
+- remedy SA collision rename ``<sa>-remedy`` when it matches the health-check
+  SA (:316-319).
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional
+
+from ..api.types import HealthCheck, PolicyRule
+from ..kube.client import EventRecorder, KubeClient
+from ..kube.errors import NotFoundError
+
+RBAC_API_VERSION = "rbac.authorization.k8s.io/v1"
+
+WF_MANAGED_BY_LABEL_KEY = "workflows.argoproj.io/managed-by"
+WF_MANAGED_BY_VALUE = "active-monitor"
+
+CLUSTER_LEVEL = "cluster"
+NAMESPACE_LEVEL = "namespace"
+
+# Least-privilege read-only rules for monitoring workflows
+# (healthcheck_controller.go:85-101).
+DEFAULT_HEALTHCHECK_RULES: List[PolicyRule] = [
+    PolicyRule(
+        api_groups=[""],
+        resources=["pods", "nodes", "events", "services", "configmaps", "namespaces", "endpoints"],
+        verbs=["get", "list", "watch"],
+    ),
+    PolicyRule(
+        api_groups=["apps"],
+        resources=["deployments", "replicasets", "statefulsets", "daemonsets"],
+        verbs=["get", "list", "watch"],
+    ),
+    PolicyRule(
+        api_groups=["argoproj.io"],
+        resources=["workflows"],
+        verbs=["get", "list", "watch"],
+    ),
+]
+
+# Scoped write rules for remedy workflows (healthcheck_controller.go:104-120).
+DEFAULT_REMEDY_RULES: List[PolicyRule] = [
+    PolicyRule(
+        api_groups=[""],
+        resources=["pods", "events", "services", "configmaps", "endpoints"],
+        verbs=["get", "list", "watch", "create", "update", "patch", "delete"],
+    ),
+    PolicyRule(
+        api_groups=["apps"],
+        resources=["deployments", "replicasets", "statefulsets"],
+        verbs=["get", "list", "watch", "create", "update", "patch", "delete"],
+    ),
+    PolicyRule(
+        api_groups=["argoproj.io"],
+        resources=["workflows"],
+        verbs=["get", "list", "watch", "create", "update", "patch", "delete"],
+    ),
+]
+
+
+def resolve_rbac_rules(
+    custom: List[PolicyRule], defaults: List[PolicyRule]
+) -> List[PolicyRule]:
+    """Custom rules win when non-empty (healthcheck_controller.go:124-129)."""
+    return custom if custom else defaults
+
+
+def _managed_labels() -> Dict[str, str]:
+    return {WF_MANAGED_BY_LABEL_KEY: WF_MANAGED_BY_VALUE}
+
+
+def _is_managed(obj: Dict[str, Any]) -> bool:
+    labels = (obj.get("metadata") or {}).get("labels") or {}
+    return labels.get(WF_MANAGED_BY_LABEL_KEY) == WF_MANAGED_BY_VALUE
+
+
+class RBACProvisioner:
+    """Creates and deletes the workflow RBAC objects through a KubeClient
+    (the reference threads a typed clientset through each helper so unit
+    tests can pass a fake — here the client itself is injectable)."""
+
+    def __init__(self, client: KubeClient, recorder: Optional[EventRecorder] = None):
+        self.client = client
+        self.recorder = recorder
+
+    async def _event(self, hc_obj: Dict[str, Any], ev_type: str, message: str) -> None:
+        if self.recorder is not None:
+            await self.recorder.event(hc_obj, ev_type, ev_type, message)
+
+    # -- creates (get-then-create, reuse if present) -----------------------
+
+    async def create_service_account(self, name: str, namespace: str) -> str:
+        try:
+            sa = await self.client.get("v1", "ServiceAccount", namespace, name)
+            return sa["metadata"]["name"]
+        except NotFoundError:
+            pass
+        sa = {
+            "apiVersion": "v1",
+            "kind": "ServiceAccount",
+            "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
+        }
+        created = await self.client.create(sa)
+        return created["metadata"]["name"]
+
+    async def create_cluster_role(self, name: str, rules: List[PolicyRule]) -> str:
+        try:
+            cr = await self.client.get(RBAC_API_VERSION, "ClusterRole", "", name)
+            return cr["metadata"]["name"]
+        except NotFoundError:
+            pass
+        cr = {
+            "apiVersion": RBAC_API_VERSION,
+            "kind": "ClusterRole",
+            "metadata": {"name": name, "labels": _managed_labels()},
+            "rules": [r.to_dict() for r in rules],
+        }
+        created = await self.client.create(cr)
+        return created["metadata"]["name"]
+
+    async def create_cluster_role_binding(
+        self, name: str, role_name: str, sa_name: str, sa_namespace: str
+    ) -> str:
+        try:
+            crb = await self.client.get(RBAC_API_VERSION, "ClusterRoleBinding", "", name)
+            return crb["metadata"]["name"]
+        except NotFoundError:
+            pass
+        crb = {
+            "apiVersion": RBAC_API_VERSION,
+            "kind": "ClusterRoleBinding",
+            "metadata": {"name": name, "labels": _managed_labels()},
+            "roleRef": {
+                "apiGroup": "rbac.authorization.k8s.io",
+                "kind": "ClusterRole",
+                "name": role_name,
+            },
+            "subjects": [
+                {"kind": "ServiceAccount", "name": sa_name, "namespace": sa_namespace}
+            ],
+        }
+        created = await self.client.create(crb)
+        return created["metadata"]["name"]
+
+    async def create_namespace_role(
+        self, name: str, namespace: str, rules: List[PolicyRule]
+    ) -> str:
+        try:
+            role = await self.client.get(RBAC_API_VERSION, "Role", namespace, name)
+            return role["metadata"]["name"]
+        except NotFoundError:
+            pass
+        role = {
+            "apiVersion": RBAC_API_VERSION,
+            "kind": "Role",
+            "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
+            "rules": [r.to_dict() for r in rules],
+        }
+        created = await self.client.create(role)
+        return created["metadata"]["name"]
+
+    async def create_namespace_role_binding(
+        self, name: str, role_name: str, sa_name: str, namespace: str
+    ) -> str:
+        try:
+            rb = await self.client.get(RBAC_API_VERSION, "RoleBinding", namespace, name)
+            return rb["metadata"]["name"]
+        except NotFoundError:
+            pass
+        rb = {
+            "apiVersion": RBAC_API_VERSION,
+            "kind": "RoleBinding",
+            "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
+            "roleRef": {
+                "apiGroup": "rbac.authorization.k8s.io",
+                "kind": "Role",
+                "name": role_name,
+            },
+            "subjects": [
+                {"kind": "ServiceAccount", "name": sa_name, "namespace": namespace}
+            ],
+        }
+        created = await self.client.create(rb)
+        return created["metadata"]["name"]
+
+    # -- deletes (only objects carrying the managed-by label) --------------
+
+    async def _delete_if_managed(self, api_version: str, kind: str, namespace: str, name: str) -> None:
+        obj = await self.client.get(api_version, kind, namespace, name)  # NotFound propagates
+        if _is_managed(obj):
+            await self.client.delete(api_version, kind, namespace, name)
+
+    async def delete_service_account(self, name: str, namespace: str) -> None:
+        await self._delete_if_managed("v1", "ServiceAccount", namespace, name)
+
+    async def delete_cluster_role(self, name: str) -> None:
+        await self._delete_if_managed(RBAC_API_VERSION, "ClusterRole", "", name)
+
+    async def delete_cluster_role_binding(self, name: str) -> None:
+        await self._delete_if_managed(RBAC_API_VERSION, "ClusterRoleBinding", "", name)
+
+    async def delete_namespace_role(self, name: str, namespace: str) -> None:
+        await self._delete_if_managed(RBAC_API_VERSION, "Role", namespace, name)
+
+    async def delete_namespace_role_binding(self, name: str, namespace: str) -> None:
+        await self._delete_if_managed(RBAC_API_VERSION, "RoleBinding", namespace, name)
+
+    # -- orchestration ------------------------------------------------------
+
+    async def create_rbac_for_workflow(self, hc: HealthCheck, workflow_type: str) -> None:
+        """``workflow_type`` ∈ {"healthCheck", "remedy"}
+        (healthcheck_controller.go:302-415). May rename the remedy SA on
+        collision — a spec mutation visible for the rest of the reconcile,
+        as in the reference (:316-319)."""
+        hc_obj = hc.to_dict()
+        level = hc.spec.level
+        hc_sa = hc.spec.workflow.resource.service_account
+        wf_namespace = hc.spec.workflow.resource.namespace
+
+        remedy_sa = ""
+        wf_remedy_namespace = ""
+        if not hc.spec.remedy_workflow.is_empty():
+            if hc.spec.remedy_workflow.resource is None:
+                await self._event(hc_obj, "Warning", "RemedyWorkflow is set but Resource is nil")
+                raise ValueError("RemedyWorkflow is set but Resource is nil")
+            if hc.spec.remedy_workflow.resource.service_account == "":
+                await self._event(
+                    hc_obj, "Warning", "ServiceAccount for the RemedyWorkflow is not specified"
+                )
+                raise ValueError("ServiceAccount for the RemedyWorkflow is not specified")
+            if hc_sa == hc.spec.remedy_workflow.resource.service_account:
+                hc.spec.remedy_workflow.resource.service_account = hc_sa + "-remedy"
+            remedy_sa = hc.spec.remedy_workflow.resource.service_account
+            wf_remedy_namespace = hc.spec.remedy_workflow.resource.namespace
+
+        is_remedy = workflow_type == "remedy"
+        if is_remedy:
+            await self.create_service_account(remedy_sa, wf_remedy_namespace)
+        else:
+            await self.create_service_account(hc_sa, wf_namespace)
+
+        hc_rules = resolve_rbac_rules(hc.spec.workflow.rbac_rules, DEFAULT_HEALTHCHECK_RULES)
+        remedy_rules = resolve_rbac_rules(hc.spec.remedy_workflow.rbac_rules, DEFAULT_REMEDY_RULES)
+
+        if level == CLUSTER_LEVEL:
+            if not is_remedy:
+                await self.create_cluster_role(hc_sa + "-cluster-role", hc_rules)
+                await self.create_cluster_role_binding(
+                    hc_sa + "-cluster-role-binding", hc_sa + "-cluster-role", hc_sa, wf_namespace
+                )
+            else:
+                await self.create_cluster_role(remedy_sa + "-cluster-role", remedy_rules)
+                await self.create_cluster_role_binding(
+                    remedy_sa + "-cluster-role-binding",
+                    remedy_sa + "-cluster-role",
+                    remedy_sa,
+                    wf_remedy_namespace,
+                )
+        elif level == NAMESPACE_LEVEL:
+            if not is_remedy:
+                await self.create_namespace_role(hc_sa + "-ns-role", wf_namespace, hc_rules)
+                await self.create_namespace_role_binding(
+                    hc_sa + "-ns-role-binding", hc_sa + "-ns-role", hc_sa, wf_namespace
+                )
+            else:
+                await self.create_namespace_role(
+                    remedy_sa + "-ns-role", wf_remedy_namespace, remedy_rules
+                )
+                await self.create_namespace_role_binding(
+                    remedy_sa + "-ns-role-binding",
+                    remedy_sa + "-ns-role",
+                    remedy_sa,
+                    wf_remedy_namespace,
+                )
+        else:
+            await self._event(hc_obj, "Warning", "level is not set")
+            raise ValueError("level is not set")
+
+    async def delete_rbac_for_workflow(self, hc: HealthCheck) -> None:
+        """Tear down the remedy RBAC after a remedy run
+        (healthcheck_controller.go:417-474). Only remedy objects are deleted,
+        and only when labeled managed-by active-monitor."""
+        if hc.spec.remedy_workflow.resource is None:
+            return
+        level = hc.spec.level
+        remedy_sa = hc.spec.remedy_workflow.resource.service_account
+        ns = hc.spec.remedy_workflow.resource.namespace
+        await self.delete_service_account(remedy_sa, ns)
+        if level == CLUSTER_LEVEL:
+            await self.delete_cluster_role(remedy_sa + "-cluster-role")
+            await self.delete_cluster_role_binding(remedy_sa + "-cluster-role-binding")
+        elif level == NAMESPACE_LEVEL:
+            await self.delete_namespace_role(remedy_sa + "-ns-role", ns)
+            await self.delete_namespace_role_binding(remedy_sa + "-ns-role-binding", ns)
+        else:
+            await self._event(hc.to_dict(), "Warning", "level is not set")
+            raise ValueError("level is not set")

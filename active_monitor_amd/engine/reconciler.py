@@ -1,0 +1,630 @@
+"""The HealthCheck reconciler — the heart of the framework.
+
+Re-implements every behavior of the reference's 1,485-line reconciler
+(internal/controllers/healthcheck_controller.go) with one deliberate
+architectural change (SURVEY.md §7 "the single biggest architectural
+decision"): the workflow watch never blocks a reconcile worker — it runs as a
+detached asyncio task — and repeat timers feed back through the workqueue, so
+``MaxConcurrentReconciles`` genuinely bounds reconcile work and repeats are
+reconcile-driven (picking up fresh spec, recomputing cron intervals every
+cycle — the reference's bare ``time.AfterFunc`` repeats re-use stale values,
+:479-500,:746-755).
+
+Behavior contracts kept bit-for-bit:
+
+- schedule branches: pause → ``Stopped`` + explanatory ErrorMessage (:238-250),
+  cron → ``RepeatAfterSec = int(next−now)+1`` (:251-263), already-scheduled
+  dedup (:264-267),
+- RBAC provision → parse → submit (ownerRef, generateName, instance-id label,
+  podGC, activeDeadlineSeconds) → watch with inverse-exponential backoff →
+  status mutation + metrics → remedy state machine → re-arm timer,
+- IEB timeout synthesizes ``{phase: Failed, message: Failed}`` (:627-632),
+- remedy RunsLimit/ResetInterval state machine with the exact RemedyStatus
+  strings and counter-zeroing sets (:649-660, :677-721),
+- remedy RBAC create → use → delete cycle (:759-786),
+- timers stopped on CR deletion (:180-184).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional, Set, Tuple
+
+from .. import API_VERSION
+from ..api.types import HealthCheck, k8s_now, parse_k8s_time
+from ..kube.client import EventRecorder, KubeClient
+from ..kube.errors import ConflictError, NotFoundError, is_storage_error
+from ..kube.registry import WF_API_VERSION, WF_KIND
+from ..metrics import (
+    MonitorError,
+    MonitorFinishedTime,
+    MonitorRuntime,
+    MonitorStartedTime,
+    MonitorSuccess,
+    create_dynamic_prometheus_metric,
+)
+from .backoff import IEBTimeoutError, InverseExponentialBackoff, compute_backoff_params
+from .cronx import CronParseError, seconds_until_next
+from .parse import (
+    parse_remedy_workflow_from_healthcheck,
+    parse_workflow_from_healthcheck,
+)
+from .rbac import RBACProvisioner
+from .workqueue import WorkQueue
+
+log = logging.getLogger("active_monitor_amd.reconciler")
+
+HC_KIND = "HealthCheck"
+SUCC_STR = "Succeeded"
+FAIL_STR = "Failed"
+REMEDY = "remedy"
+HEALTHCHECK = "healthCheck"
+TIMER_FLAG = "timer"
+
+
+@dataclass
+class ReconcileResult:
+    requeue_after: float = 0.0
+    error: Optional[BaseException] = None
+
+
+@dataclass
+class RepeatTimer:
+    """Entry in RepeatTimersByName. Like the reference's expired ``time.Timer``
+    values (:746-755), a fired entry stays in the map so the already-scheduled
+    dedup check keeps holding during the run it triggered."""
+
+    handle: Optional[asyncio.TimerHandle] = None
+    fired: bool = False
+
+    def stop(self) -> None:
+        if self.handle is not None:
+            self.handle.cancel()
+
+
+class HealthCheckReconciler:
+    def __init__(
+        self,
+        client: KubeClient,
+        recorder: EventRecorder,
+        max_parallel: int = 10,
+        queue: Optional[WorkQueue] = None,
+    ):
+        self.client = client
+        self.recorder = recorder
+        self.max_parallel = max_parallel
+        # NB: not `queue or WorkQueue()` — an empty WorkQueue is falsy (__len__)
+        self.queue = queue if queue is not None else WorkQueue()
+        self.rbac = RBACProvisioner(client, recorder)
+        self.repeat_timers_by_name: Dict[str, RepeatTimer] = {}
+        self._watch_tasks: Dict[str, Set[asyncio.Task]] = {}
+        # observability for benchmarks/tests
+        self.reconcile_count = 0
+        self.completed_runs = 0
+
+    # ------------------------------------------------------------------
+    # timers
+    # ------------------------------------------------------------------
+
+    def get_timer_by_name(self, name: str) -> Optional[RepeatTimer]:
+        return self.repeat_timers_by_name.get(name)
+
+    def _arm_repeat_timer(self, name: str, namespace: str, delay: float) -> None:
+        old = self.repeat_timers_by_name.get(name)
+        if old is not None:
+            old.stop()
+        entry = RepeatTimer()
+        loop = asyncio.get_running_loop()
+        entry.handle = loop.call_later(
+            max(0.0, delay), self._on_timer_fired, entry, name, namespace
+        )
+        self.repeat_timers_by_name[name] = entry
+
+    def _on_timer_fired(self, entry: RepeatTimer, name: str, namespace: str) -> None:
+        entry.fired = True
+        # repeats flow through the workqueue (bounded by max_parallel); the
+        # TIMER flag bypasses only the already-scheduled dedup branch.
+        # Synchronous enqueue — no fire-and-forget task to lose.
+        self.queue.add_nowait((namespace, name), {TIMER_FLAG})
+
+    def _stop_timer(self, name: str) -> bool:
+        entry = self.repeat_timers_by_name.pop(name, None)
+        if entry is not None:
+            entry.stop()
+            return True
+        return False
+
+    # ------------------------------------------------------------------
+    # watch-task registry
+    # ------------------------------------------------------------------
+
+    def _spawn_watch(self, name: str, coro) -> asyncio.Task:
+        task = asyncio.get_running_loop().create_task(coro)
+        self._watch_tasks.setdefault(name, set()).add(task)
+
+        def _done(t: asyncio.Task, name=name) -> None:
+            tasks = self._watch_tasks.get(name)
+            if tasks is not None:
+                tasks.discard(t)
+                if not tasks:
+                    self._watch_tasks.pop(name, None)
+            if not t.cancelled() and t.exception() is not None:
+                log.error("watch task for %s failed: %s", name, t.exception())
+
+        task.add_done_callback(_done)
+        return task
+
+    def _cancel_watches(self, name: str) -> None:
+        for t in list(self._watch_tasks.get(name, ())):
+            t.cancel()
+
+    def active_watches(self) -> int:
+        return sum(len(v) for v in self._watch_tasks.values())
+
+    async def drain(self, timeout: float = 30.0) -> None:
+        """Await all in-flight watch tasks (test/shutdown helper)."""
+        deadline = time.monotonic() + timeout
+        while self._watch_tasks and time.monotonic() < deadline:
+            tasks = [t for ts in self._watch_tasks.values() for t in ts]
+            if not tasks:
+                break
+            await asyncio.wait(tasks, timeout=min(1.0, deadline - time.monotonic()))
+
+    # ------------------------------------------------------------------
+    # events
+    # ------------------------------------------------------------------
+
+    async def _event(self, hc: HealthCheck, ev_type: str, message: str) -> None:
+        await self.recorder.event(hc.to_dict(), ev_type, ev_type, message)
+
+    # ------------------------------------------------------------------
+    # Reconcile entry (reference :170-223)
+    # ------------------------------------------------------------------
+
+    async def reconcile(
+        self, namespace: str, name: str, flags: Optional[Set[str]] = None
+    ) -> ReconcileResult:
+        flags = flags or set()
+        self.reconcile_count += 1
+        try:
+            obj = await self.client.get(API_VERSION, HC_KIND, namespace, name)
+        except NotFoundError:
+            # CR deleted: stop the repeat timer so self-scheduling halts
+            # (:180-184); in-flight watches are cancelled proactively (the
+            # reference lets them die on workflow NotFound after GC).
+            if self._stop_timer(name):
+                log.info("cancelled rescheduled workflow for deleted healthcheck %s", name)
+            self._cancel_watches(name)
+            return ReconcileResult()
+        hc = HealthCheck.from_dict(obj)
+        return await self._process_or_recover(hc, from_timer=TIMER_FLAG in flags)
+
+    async def _process_or_recover(self, hc: HealthCheck, from_timer: bool) -> ReconcileResult:
+        """Panic guard + error policy (reference :190-223). The reference's
+        trailing full-object Update (:208-215) is omitted: with the status
+        subresource enabled it can persist neither status nor the in-memory
+        spec mutations of the fresh object it writes — a pure no-op write per
+        reconcile. Status persistence happens where it matters, in the watch
+        completion paths (:734,:858) and the pause branch (:246)."""
+        try:
+            return await self._process_healthcheck(hc, from_timer)
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:  # recover() equivalent — never crash a worker
+            if is_storage_error(e):
+                return ReconcileResult()
+            log.warning("healthcheck %s process error: %s", hc.name, e)
+            return ReconcileResult(requeue_after=1.0, error=e)
+
+    # ------------------------------------------------------------------
+    # Scheduler (reference :225-291)
+    # ------------------------------------------------------------------
+
+    async def _process_healthcheck(self, hc: HealthCheck, from_timer: bool) -> ReconcileResult:
+        spec = hc.spec
+        if spec.workflow.resource is None:
+            return ReconcileResult()
+        wf_namespace = spec.workflow.resource.namespace
+
+        finished_unix = 0.0
+        t = parse_k8s_time(hc.status.finished_at)
+        if t is not None:
+            finished_unix = t.timestamp()
+
+        if spec.repeat_after_sec <= 0 and spec.schedule.cron == "":
+            # pause branch (:238-250) — exact status strings
+            hc.status.status = "Stopped"
+            hc.status.error_message = (
+                "workflow execution is stopped; either spec.RepeatAfterSec or "
+                "spec.Schedule must be provided. spec.RepeatAfterSec set to "
+                f"{spec.repeat_after_sec}. spec.Schedule set to {spec.schedule.go_string()}"
+            )
+            hc.status.finished_at = k8s_now()
+            await self._event(
+                hc,
+                "Warning",
+                "Workflow execution is stopped; either spec.RepeatAfterSec or "
+                "spec.Schedule must be provided",
+            )
+            await self.update_healthcheck_status(hc)
+            return ReconcileResult()
+        elif not from_timer and self._watch_tasks.get(hc.name):
+            # a run for this CR is already in flight: spurious reconciles
+            # (informer list/watch overlap at startup, spec edits mid-run)
+            # must not submit a duplicate workflow. The reference double-
+            # submits here and parks a second blocked goroutine per duplicate
+            # (SURVEY.md §2.3.1); the clean rebuild closes that hole.
+            return ReconcileResult()
+        elif spec.repeat_after_sec <= 0 and spec.schedule.cron != "":
+            # cron branch (:251-263): +1s compensates integer truncation
+            try:
+                hc.spec.repeat_after_sec = seconds_until_next(spec.schedule.cron)
+            except CronParseError as e:
+                await self._event(hc, "Warning", "Fail to parse cron")
+                raise e
+        elif (
+            not from_timer
+            and int(time.time() - finished_unix) < spec.repeat_after_sec
+            and self.get_timer_by_name(hc.name) is not None
+        ):
+            # already executed recently and a repeat is scheduled (:264-267)
+            return ReconcileResult()
+
+        try:
+            await self.rbac.create_rbac_for_workflow(hc, HEALTHCHECK)
+        except Exception as e:
+            await self._event(hc, "Warning", "Error creating RBAC for HealthCheckWorkflow")
+            raise e
+
+        generated_name = await self.create_submit_workflow(hc)
+        # non-blocking watch: the reconcile worker is freed immediately
+        self._spawn_watch(
+            hc.name, self.watch_workflow_reschedule(wf_namespace, generated_name, hc)
+        )
+        return ReconcileResult()
+
+    # ------------------------------------------------------------------
+    # Submit (reference :502-571)
+    # ------------------------------------------------------------------
+
+    def _owner_reference(self, hc: HealthCheck) -> Dict[str, Any]:
+        return {
+            "kind": HC_KIND,
+            "apiVersion": API_VERSION,
+            "name": hc.name,
+            "uid": hc.metadata.uid,
+            "controller": True,
+        }
+
+    async def create_submit_workflow(self, hc: HealthCheck) -> str:
+        try:
+            spec, labels = parse_workflow_from_healthcheck(hc)
+        except Exception as e:
+            await self._event(hc, "Warning", "Error creating or submitting workflow")
+            raise e
+        wf = {
+            "apiVersion": WF_API_VERSION,
+            "kind": WF_KIND,
+            "metadata": {
+                "generateName": hc.spec.workflow.generate_name,
+                "namespace": hc.spec.workflow.resource.namespace,
+                "labels": labels,
+                "ownerReferences": [self._owner_reference(hc)],
+            },
+            "spec": spec,
+        }
+        created = await self.client.create(wf)
+        await self._event(hc, "Normal", "Successfully created workflow")
+        return created["metadata"]["name"]
+
+    async def create_submit_remedy_workflow(self, hc: HealthCheck) -> str:
+        if hc.spec.remedy_workflow.resource is None:
+            raise ValueError("RemedyWorkflow Resource is nil")
+        try:
+            spec, labels = parse_remedy_workflow_from_healthcheck(hc)
+        except Exception as e:
+            await self._event(hc, "Warning", "Error creating or submitting remedyworkflow")
+            raise e
+        wf = {
+            "apiVersion": WF_API_VERSION,
+            "kind": WF_KIND,
+            "metadata": {
+                "generateName": hc.spec.remedy_workflow.generate_name,
+                "namespace": hc.spec.remedy_workflow.resource.namespace,
+                "labels": labels,
+                "ownerReferences": [self._owner_reference(hc)],
+            },
+            "spec": spec,
+        }
+        created = await self.client.create(wf)
+        await self._event(hc, "Normal", "Successfully created remedyWorkflow")
+        return created["metadata"]["name"]
+
+    # ------------------------------------------------------------------
+    # Watch: health-check workflow (reference :607-757)
+    # ------------------------------------------------------------------
+
+    async def _poll_workflow(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
+        wf = await self.client.get(WF_API_VERSION, WF_KIND, namespace, name)
+        status = wf.get("status")
+        return status if isinstance(status, dict) else None
+
+    async def watch_workflow_reschedule(
+        self, wf_namespace: str, wf_name: str, hc: HealthCheck
+    ) -> None:
+        then = k8s_now()
+        then_unix = time.time()
+        repeat_after_sec = hc.spec.repeat_after_sec
+        max_t, min_t, factor, timeout = compute_backoff_params(
+            hc.spec.backoff_max, hc.spec.backoff_min, hc.spec.backoff_factor,
+            hc.spec.workflow.timeout,
+        )
+        ieb: Optional[InverseExponentialBackoff] = None
+        timed_out = False
+        try:
+            ieb = InverseExponentialBackoff(max_t, min_t, timeout, factor)
+        except ValueError:
+            # invalid params (e.g. timeout 0): the reference's loop sees a
+            # constructor error on its first iteration and synthesizes Failed
+            timed_out = True
+
+        first = True
+        while True:
+            if not first:
+                try:
+                    await ieb.next()
+                except IEBTimeoutError:
+                    timed_out = True
+            first = False
+            try:
+                status = await self._poll_workflow(wf_namespace, wf_name)
+            except NotFoundError:
+                # parent healthcheck likely deleted; don't reschedule (:618-622)
+                await self._event(
+                    hc,
+                    "Warning",
+                    "Error attempting to find workflow for healthcheck. This may "
+                    "indicate that either the healthcheck was removed or the Workflow "
+                    "was GC'd before active-monitor could obtain the status",
+                )
+                return
+            if timed_out:
+                status = {"phase": FAIL_STR, "message": FAIL_STR}
+                await self._event(hc, "Warning", "Workflow timed out")
+            if status is not None:
+                phase = status.get("phase")
+                now = k8s_now()
+                now_unix = time.time()
+                if phase == SUCC_STR:
+                    await self._event(hc, "Normal", "Workflow status is Succeeded")
+                    hc.status.status = SUCC_STR
+                    hc.status.started_at = then
+                    hc.status.finished_at = now
+                    hc.status.success_count += 1
+                    hc.status.total_healthcheck_runs = (
+                        hc.status.success_count + hc.status.failed_count
+                    )
+                    hc.status.last_successful_workflow = wf_name
+                    MonitorSuccess.labels(hc.name, HEALTHCHECK).inc()
+                    MonitorRuntime.labels(hc.name, HEALTHCHECK).set(now_unix - then_unix)
+                    MonitorStartedTime.labels(hc.name, HEALTHCHECK).set(int(then_unix))
+                    MonitorFinishedTime.labels(hc.name, HEALTHCHECK).set(int(now_unix))
+                    # custom metrics from workflow output parameters — the
+                    # reference documents this (README.md:275-285) but never
+                    # wires it; here the documented feature is real.
+                    create_dynamic_prometheus_metric(hc.name, status)
+                    if (
+                        not hc.spec.remedy_workflow.is_empty()
+                        and hc.status.remedy_total_runs >= 1
+                    ):
+                        hc.status.reset_remedy()
+                        hc.status.remedy_status = "HealthCheck Passed so Remedy is reset"
+                        await self._event(hc, "Normal", "HealthCheck passed so Remedy is reset")
+                    break
+                elif phase == FAIL_STR:
+                    await self._event(hc, "Warning", "Workflow status is Failed")
+                    hc.status.status = FAIL_STR
+                    hc.status.started_at = then
+                    hc.status.finished_at = now
+                    hc.status.last_failed_at = now
+                    msg = status.get("message")
+                    hc.status.error_message = msg if isinstance(msg, str) else ""
+                    hc.status.failed_count += 1
+                    hc.status.total_healthcheck_runs = (
+                        hc.status.success_count + hc.status.failed_count
+                    )
+                    hc.status.last_failed_workflow = wf_name
+                    MonitorError.labels(hc.name, HEALTHCHECK).inc()
+                    MonitorStartedTime.labels(hc.name, HEALTHCHECK).set(int(then_unix))
+                    MonitorFinishedTime.labels(hc.name, HEALTHCHECK).set(int(now_unix))
+                    await self._maybe_run_remedy(hc, now_unix)
+                    break
+
+        await self._finish_and_reschedule(hc, wf_namespace, wf_name, repeat_after_sec)
+        self.completed_runs += 1
+
+    async def _maybe_run_remedy(self, hc: HealthCheck, now_unix: float) -> None:
+        """The remedy trigger state machine (reference :677-721)."""
+        if hc.spec.remedy_workflow.is_empty():
+            return
+        limit = hc.spec.remedy_runs_limit
+        reset_interval = hc.spec.remedy_reset_interval
+        if limit != 0 and reset_interval != 0:
+            if limit > hc.status.remedy_total_runs:
+                await self.process_remedy_workflow(hc)
+            else:
+                rf = parse_k8s_time(hc.status.remedy_finished_at)
+                since_last = int(now_unix - rf.timestamp()) if rf is not None else reset_interval + 1
+                if reset_interval >= since_last:
+                    log.info(
+                        "skipping remedy for %s: remedy limit met, will run after reset interval",
+                        hc.name,
+                    )
+                else:
+                    hc.status.reset_remedy()
+                    hc.status.remedy_status = "RemedyResetInterval elapsed so Remedy is reset"
+                    await self._event(
+                        hc, "Normal", "RemedyResetInterval elapsed so Remedy is reset"
+                    )
+                    await self.process_remedy_workflow(hc)
+        else:
+            await self.process_remedy_workflow(hc)
+
+    async def _finish_and_reschedule(
+        self, hc: HealthCheck, wf_namespace: str, wf_name: str, repeat_after_sec: float
+    ) -> None:
+        """Persist status and re-arm the repeat timer (reference :729-757).
+
+        The timer is armed BEFORE the status write so the MODIFIED event the
+        write triggers always observes an armed timer in the dedup branch —
+        closing a duplicate-submission race the reference leaves open by
+        updating first (:734) and arming after (:746)."""
+        try:
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+        except NotFoundError:
+            return
+        if (fresh.get("metadata") or {}).get("deletionTimestamp"):
+            return
+        self._arm_repeat_timer(hc.name, hc.namespace, repeat_after_sec)
+        try:
+            await self.update_healthcheck_status(hc)
+        except NotFoundError:
+            self._stop_timer(hc.name)
+            return
+        except Exception as e:
+            await self._event(hc, "Warning", "Error updating healthcheck resource")
+            self._stop_timer(hc.name)
+            raise e
+        await self._event(hc, "Normal", "Rescheduled workflow for next run")
+
+    # ------------------------------------------------------------------
+    # Remedy (reference :759-874)
+    # ------------------------------------------------------------------
+
+    async def process_remedy_workflow(self, hc: HealthCheck) -> None:
+        """create remedy RBAC → submit → watch → delete remedy RBAC
+        (reference :759-786)."""
+        await self.rbac.create_rbac_for_workflow(hc, REMEDY)
+        generated_name = await self.create_submit_remedy_workflow(hc)
+        # watch in the remedy's own namespace — the reference watches in the
+        # health-check workflow's namespace (:773), which mis-targets when the
+        # two differ; fixed here with no CR-visible change for the common case
+        remedy_ns = hc.spec.remedy_workflow.resource.namespace
+        await self.watch_remedy_workflow(remedy_ns, generated_name, hc)
+        await self.rbac.delete_rbac_for_workflow(hc)
+
+    async def watch_remedy_workflow(
+        self, wf_namespace: str, wf_name: str, hc: HealthCheck
+    ) -> None:
+        """Remedy watch loop (reference :788-874): backoff params derive from
+        the *health-check* workflow's timeout with factor fixed at 0.5
+        (:791-801), a reference quirk kept for behavioral parity."""
+        then = k8s_now()
+        then_unix = time.time()
+        max_t, min_t, _factor, timeout = compute_backoff_params(
+            0, 0, "", hc.spec.workflow.timeout
+        )
+        ieb: Optional[InverseExponentialBackoff] = None
+        timed_out = False
+        try:
+            ieb = InverseExponentialBackoff(max_t, min_t, timeout, 0.5)
+        except ValueError:
+            timed_out = True
+
+        first = True
+        while True:
+            if not first:
+                try:
+                    await ieb.next()
+                except IEBTimeoutError:
+                    timed_out = True
+            first = False
+            try:
+                status = await self._poll_workflow(wf_namespace, wf_name)
+            except NotFoundError:
+                return
+            if timed_out:
+                status = {"phase": FAIL_STR, "message": FAIL_STR}
+                await self._event(hc, "Warning", "remedy workflow is timedout")
+            if status is not None:
+                phase = status.get("phase")
+                now = k8s_now()
+                now_unix = time.time()
+                if phase == SUCC_STR:
+                    await self._event(hc, "Normal", "Remedy workflow status is Succeeded")
+                    hc.status.remedy_status = SUCC_STR
+                    hc.status.remedy_started_at = then
+                    hc.status.remedy_finished_at = now
+                    hc.status.remedy_success_count += 1
+                    hc.status.remedy_total_runs = (
+                        hc.status.remedy_success_count + hc.status.remedy_failed_count
+                    )
+                    # reference quirk kept: remedy watch writes the shared
+                    # last*Workflow fields with the remedy name (:830)
+                    hc.status.last_successful_workflow = wf_name
+                    MonitorSuccess.labels(hc.name, REMEDY).inc()
+                    MonitorRuntime.labels(hc.name, REMEDY).set(now_unix - then_unix)
+                    MonitorStartedTime.labels(hc.name, REMEDY).set(int(then_unix))
+                    # reference quirk kept: finished-time gauge carries the
+                    # HEALTH-CHECK finish epoch, not the remedy's (:834)
+                    hc_fin = parse_k8s_time(hc.status.finished_at)
+                    MonitorFinishedTime.labels(hc.name, REMEDY).set(
+                        int(hc_fin.timestamp()) if hc_fin else int(now_unix)
+                    )
+                    break
+                elif phase == FAIL_STR:
+                    await self._event(hc, "Warning", "remedy workflow status is failed")
+                    hc.status.remedy_status = FAIL_STR
+                    hc.status.remedy_started_at = then
+                    hc.status.remedy_finished_at = now
+                    hc.status.remedy_last_failed_at = now
+                    msg = status.get("message")
+                    hc.status.remedy_error_message = msg if isinstance(msg, str) else ""
+                    hc.status.remedy_failed_count += 1
+                    hc.status.remedy_total_runs = (
+                        hc.status.remedy_success_count + hc.status.remedy_failed_count
+                    )
+                    hc.status.last_failed_workflow = wf_name
+                    MonitorError.labels(hc.name, REMEDY).inc()
+                    MonitorStartedTime.labels(hc.name, REMEDY).set(int(then_unix))
+                    MonitorFinishedTime.labels(hc.name, REMEDY).set(int(now_unix))
+                    break
+
+        # persist remedy status promptly (reference :856-871)
+        try:
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+        except NotFoundError:
+            return
+        if (fresh.get("metadata") or {}).get("deletionTimestamp"):
+            return
+        await self.update_healthcheck_status(hc)
+
+    # ------------------------------------------------------------------
+    # Status persistence (reference :1445-1462)
+    # ------------------------------------------------------------------
+
+    async def update_healthcheck_status(self, hc: HealthCheck, retries: int = 5) -> None:
+        """Fresh Get + status-subresource update with conflict retry."""
+        last: Optional[BaseException] = None
+        for attempt in range(retries):
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+            fresh["status"] = hc.status.to_dict()
+            try:
+                await self.client.update_status(fresh)
+                return
+            except ConflictError as e:
+                last = e
+                await asyncio.sleep(0.01 * (attempt + 1))
+        raise last if last else RuntimeError("status update failed")
+
+    # ------------------------------------------------------------------
+    # Shutdown
+    # ------------------------------------------------------------------
+
+    def stop_all(self) -> None:
+        for name in list(self.repeat_timers_by_name):
+            self._stop_timer(name)
+        for name in list(self._watch_tasks):
+            self._cancel_watches(name)

@@ -239,6 +239,12 @@ class MemoryApiServer:
                     obj["status"] = copy.deepcopy(existing["status"])
                 else:
                     obj.pop("status", None)
+            # no-op updates don't bump the resourceVersion or emit watch
+            # events (apiserver semantics — prevents self-triggering loops)
+            meta["resourceVersion"] = ex_meta.get("resourceVersion")
+            meta["generation"] = ex_meta.get("generation", 1)
+            if obj == existing:
+                return copy.deepcopy(existing)
             if obj.get("spec") != existing.get("spec"):
                 meta["generation"] = int(ex_meta.get("generation", 1)) + 1
             else:
@@ -277,6 +283,8 @@ class MemoryApiServer:
                 updated["status"] = copy.deepcopy(obj["status"])
             else:
                 updated.pop("status", None)
+            if updated == existing:  # no-op status write (apiserver semantics)
+                return copy.deepcopy(existing)
             updated["metadata"]["resourceVersion"] = self._next_rv()
             self._objects[key] = updated
             out = copy.deepcopy(updated)
