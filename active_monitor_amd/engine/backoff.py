@@ -58,20 +58,26 @@ class InverseExponentialBackoff:
     def remaining(self) -> float:
         return self.timeout - (self._clock() - self._start)
 
+    def peek_interval(self) -> float:
+        """The next wait duration, capped at the remaining budget (no decay)."""
+        return min(self._interval, max(0.0, self.remaining()))
+
+    def decay(self) -> None:
+        self._interval = max(self.min_interval, self._interval * self.factor)
+
+    def check_deadline(self) -> None:
+        if self.remaining() <= 0:
+            raise IEBTimeoutError(f"timeout of {self.timeout}s exceeded")
+
     async def next(self) -> None:
         """Sleep the current interval (capped at the remaining budget), decay
         it toward ``min_interval``, and raise :class:`IEBTimeoutError` once the
         deadline has passed."""
-        remaining = self.remaining()
-        if remaining <= 0:
-            raise IEBTimeoutError(
-                f"timeout of {self.timeout}s exceeded"
-            )
-        interval = min(self._interval, remaining)
-        self._interval = max(self.min_interval, self._interval * self.factor)
+        self.check_deadline()
+        interval = self.peek_interval()
+        self.decay()
         await self._sleep(interval)
-        if self.remaining() <= 0:
-            raise IEBTimeoutError(f"timeout of {self.timeout}s exceeded")
+        self.check_deadline()
 
 
 def compute_backoff_params(

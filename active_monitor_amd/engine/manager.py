@@ -87,6 +87,12 @@ class Manager:
                 self, health=self.health_addr, metrics=self.metrics_addr
             )
 
+        from .watchhub import WorkflowWatchHub
+
+        self.wf_hub = WorkflowWatchHub(self.client, self.namespace)
+        await self.wf_hub.start()
+        self.reconciler.wf_hub = self.wf_hub
+
         self._tasks.append(asyncio.ensure_future(self._informer()))
         for i in range(self.max_workers):
             self._tasks.append(asyncio.ensure_future(self._worker(i)))
@@ -103,6 +109,8 @@ class Manager:
         self._stopped = True
         if self._sub is not None:
             self._sub.close()
+        if getattr(self, "wf_hub", None) is not None:
+            await self.wf_hub.stop()
         await self.queue.shutdown()
         self.reconciler.stop_all()
         for t in self._tasks:

@@ -98,6 +98,9 @@ class HealthCheckReconciler:
         # NB: not `queue or WorkQueue()` — an empty WorkQueue is falsy (__len__)
         self.queue = queue if queue is not None else WorkQueue()
         self.rbac = RBACProvisioner(client, recorder)
+        # optional WorkflowWatchHub (set by the Manager): event-driven wakeups
+        # for workflow completion; None falls back to pure IEB polling
+        self.wf_hub = None
         self.repeat_timers_by_name: Dict[str, RepeatTimer] = {}
         self._watch_tasks: Dict[str, Set[asyncio.Task]] = {}
         # observability for benchmarks/tests
@@ -351,6 +354,23 @@ class HealthCheckReconciler:
         status = wf.get("status")
         return status if isinstance(status, dict) else None
 
+    async def _wait_next_poll(
+        self, ieb: InverseExponentialBackoff, namespace: str, name: str, since: Optional[int]
+    ) -> None:
+        """Wait until the next poll is due. With a watch hub the wait ends the
+        moment the workflow changes (ms-scale completion detection); without
+        one this is exactly the reference's IEB sleep. Raises IEBTimeoutError
+        past the deadline either way — the synthesized-failure semantics are
+        identical."""
+        if self.wf_hub is None:
+            await ieb.next()
+            return
+        ieb.check_deadline()
+        interval = ieb.peek_interval()
+        ieb.decay()
+        await self.wf_hub.wait_change(namespace, name, interval, since=since)
+        ieb.check_deadline()
+
     async def watch_workflow_reschedule(
         self, wf_namespace: str, wf_name: str, hc: HealthCheck
     ) -> None:
@@ -370,14 +390,10 @@ class HealthCheckReconciler:
             # constructor error on its first iteration and synthesizes Failed
             timed_out = True
 
-        first = True
         while True:
-            if not first:
-                try:
-                    await ieb.next()
-                except IEBTimeoutError:
-                    timed_out = True
-            first = False
+            # pre-poll change counter: any event landing between this poll and
+            # the wait below is detected immediately (no lost wakeups)
+            seq = self.wf_hub.seq(wf_namespace, wf_name) if self.wf_hub else None
             try:
                 status = await self._poll_workflow(wf_namespace, wf_name)
             except NotFoundError:
@@ -441,6 +457,11 @@ class HealthCheckReconciler:
                     MonitorFinishedTime.labels(hc.name, HEALTHCHECK).set(int(now_unix))
                     await self._maybe_run_remedy(hc, now_unix)
                     break
+            # not terminal yet: wait for the next poll (hub-accelerated)
+            try:
+                await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
+            except IEBTimeoutError:
+                timed_out = True
 
         await self._finish_and_reschedule(hc, wf_namespace, wf_name, repeat_after_sec)
         self.completed_runs += 1
@@ -533,14 +554,8 @@ class HealthCheckReconciler:
         except ValueError:
             timed_out = True
 
-        first = True
         while True:
-            if not first:
-                try:
-                    await ieb.next()
-                except IEBTimeoutError:
-                    timed_out = True
-            first = False
+            seq = self.wf_hub.seq(wf_namespace, wf_name) if self.wf_hub else None
             try:
                 status = await self._poll_workflow(wf_namespace, wf_name)
             except NotFoundError:
@@ -591,6 +606,10 @@ class HealthCheckReconciler:
                     MonitorStartedTime.labels(hc.name, REMEDY).set(int(then_unix))
                     MonitorFinishedTime.labels(hc.name, REMEDY).set(int(now_unix))
                     break
+            try:
+                await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
+            except IEBTimeoutError:
+                timed_out = True
 
         # persist remedy status promptly (reference :856-871)
         try:

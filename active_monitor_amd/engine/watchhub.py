@@ -1,0 +1,95 @@
+"""Workflow watch hub: event-driven completion discovery.
+
+The reference discovers workflow completion purely by polling the apiserver on
+an inverse-exponential cadence (healthcheck_controller.go:613-624), so the
+completion-detection latency is O(poll interval) — up to Timeout/2 seconds.
+This hub subscribes once to Workflow watch events and wakes any interested
+watcher the moment its workflow changes, collapsing that latency to
+milliseconds. The IEB poll remains as the fallback cadence and still governs
+the synthesized-failure deadline, so CR-visible semantics are unchanged — the
+hub is purely a wake accelerator ("make the watch non-blocking
+(watch/informer on Workflow phase or short resumable requeues)").
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import Dict, List, Optional, Tuple
+
+from ..kube.client import KubeClient
+from ..kube.registry import WF_API_VERSION, WF_KIND
+
+Key = Tuple[str, str]  # (namespace, name)
+
+
+class WorkflowWatchHub:
+    def __init__(self, client: KubeClient, namespace: Optional[str] = None):
+        self.client = client
+        self.namespace = namespace
+        self._waiters: Dict[Key, List[asyncio.Future]] = {}
+        # per-key monotonic change counter: lets a watcher detect events that
+        # fired between its poll and its wait registration (no lost wakeups)
+        self._seq: Dict[Key, int] = {}
+        self._sub = None
+        self._task: Optional[asyncio.Task] = None
+
+    def seq(self, namespace: str, name: str) -> int:
+        return self._seq.get((namespace, name), 0)
+
+    async def start(self) -> None:
+        self._sub = self.client.watch(WF_API_VERSION, WF_KIND, self.namespace)
+        self._task = asyncio.ensure_future(self._consume())
+
+    async def stop(self) -> None:
+        if self._sub is not None:
+            self._sub.close()
+        if self._task is not None:
+            self._task.cancel()
+            try:
+                await self._task
+            except (asyncio.CancelledError, Exception):
+                pass
+        for waiters in self._waiters.values():
+            for fut in waiters:
+                if not fut.done():
+                    fut.set_result(None)
+        self._waiters.clear()
+
+    async def _consume(self) -> None:
+        async for ev in self._sub:
+            meta = ev["object"].get("metadata") or {}
+            key = (meta.get("namespace", ""), meta.get("name", ""))
+            waiters = self._waiters.pop(key, ())
+            if ev["type"] == "DELETED":
+                self._seq.pop(key, None)  # prune; NotFound ends the watchers
+            else:
+                self._seq[key] = self._seq.get(key, 0) + 1
+            for fut in waiters:  # wake everyone watching it
+                if not fut.done():
+                    fut.set_result(ev["type"])
+
+    async def wait_change(
+        self, namespace: str, name: str, timeout: float, since: Optional[int] = None
+    ) -> Optional[str]:
+        """Block until the workflow changes (returns the event type) or the
+        timeout elapses (returns None). Callers re-poll the object either way,
+        so a missed event only costs one poll interval. Pass ``since`` (a value
+        from :meth:`seq` captured before the caller's poll) to return
+        immediately when a change already landed in between."""
+        key = (namespace, name)
+        if since is not None and self._seq.get(key, 0) > since:
+            return "CHANGED"
+        fut = asyncio.get_running_loop().create_future()
+        self._waiters.setdefault(key, []).append(fut)
+        try:
+            return await asyncio.wait_for(fut, timeout)
+        except asyncio.TimeoutError:
+            return None
+        finally:
+            waiters = self._waiters.get(key)
+            if waiters is not None:
+                try:
+                    waiters.remove(fut)
+                except ValueError:
+                    pass
+                if not waiters:
+                    self._waiters.pop(key, None)
