@@ -51,15 +51,32 @@ def _extract_labels(data: Dict[str, Any]) -> Dict[str, str]:
     return labels
 
 
+# libyaml loader when available (~10x faster than the pure-Python one)
+_Loader = getattr(yaml, "CSafeLoader", yaml.SafeLoader)
+
+# Workflow definitions are re-parsed every cycle (reference :894); fleets
+# re-submit the same YAML thousands of times, so cache parses by content.
+_PARSE_CACHE: Dict[bytes, Dict[str, Any]] = {}
+_PARSE_CACHE_MAX = 2048
+
+
 def _decode(content: bytes, remedy: bool) -> Dict[str, Any]:
+    from ..utils.fastcopy import deep_copy
+
+    cached = _PARSE_CACHE.get(content)
+    if cached is not None:
+        return deep_copy(cached)
     try:
-        data = yaml.safe_load(content)
+        data = yaml.load(content, Loader=_Loader)
     except yaml.YAMLError as e:
         raise WorkflowParseError(f"Invalid spec file passed: {e}") from e
     if data is None:
         data = {}
     if not isinstance(data, dict):
         raise WorkflowParseError("Invalid spec file passed: not a mapping")
+    if len(_PARSE_CACHE) >= _PARSE_CACHE_MAX:
+        _PARSE_CACHE.clear()  # simple bound; refill costs one parse per doc
+    _PARSE_CACHE[content] = deep_copy(data)
     return data
 
 

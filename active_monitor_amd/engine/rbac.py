@@ -198,9 +198,19 @@ class RBACProvisioner:
     # -- deletes (only objects carrying the managed-by label) --------------
 
     async def _delete_if_managed(self, api_version: str, kind: str, namespace: str, name: str) -> None:
-        obj = await self.client.get(api_version, kind, namespace, name)  # NotFound propagates
+        # Already-gone objects count as deleted. (The reference propagates the
+        # Get error here (:1164-1166), which makes concurrent remedy teardowns
+        # of a shared SA fail each other — an availability fix, not a
+        # semantic change.)
+        try:
+            obj = await self.client.get(api_version, kind, namespace, name)
+        except NotFoundError:
+            return
         if _is_managed(obj):
-            await self.client.delete(api_version, kind, namespace, name)
+            try:
+                await self.client.delete(api_version, kind, namespace, name)
+            except NotFoundError:
+                return
 
     async def delete_service_account(self, name: str, namespace: str) -> None:
         await self._delete_if_managed("v1", "ServiceAccount", namespace, name)

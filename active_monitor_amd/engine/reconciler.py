@@ -455,7 +455,16 @@ class HealthCheckReconciler:
                     MonitorError.labels(hc.name, HEALTHCHECK).inc()
                     MonitorStartedTime.labels(hc.name, HEALTHCHECK).set(int(then_unix))
                     MonitorFinishedTime.labels(hc.name, HEALTHCHECK).set(int(now_unix))
-                    await self._maybe_run_remedy(hc, now_unix)
+                    try:
+                        await self._maybe_run_remedy(hc, now_unix)
+                    except asyncio.CancelledError:
+                        raise
+                    except Exception as e:
+                        # a remedy failure must not lose the health-check
+                        # status or the repeat schedule (the reference aborts
+                        # the whole watch here, stalling the CR — :686-688)
+                        log.warning("remedy for %s failed: %s", hc.name, e)
+                        await self._event(hc, "Warning", "Error executing RemedyWorkflow")
                     break
             # not terminal yet: wait for the next poll (hub-accelerated)
             try:
@@ -463,8 +472,10 @@ class HealthCheckReconciler:
             except IEBTimeoutError:
                 timed_out = True
 
-        await self._finish_and_reschedule(hc, wf_namespace, wf_name, repeat_after_sec)
-        self.completed_runs += 1
+        try:
+            await self._finish_and_reschedule(hc, wf_namespace, wf_name, repeat_after_sec)
+        finally:
+            self.completed_runs += 1
 
     async def _maybe_run_remedy(self, hc: HealthCheck, now_unix: float) -> None:
         """The remedy trigger state machine (reference :677-721)."""

@@ -21,13 +21,13 @@ The core is synchronous and thread-safe; watches are asyncio queues fed via
 from __future__ import annotations
 
 import asyncio
-import copy
 import random
-import string
 import threading
+from collections import deque
 from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
 
 from ..api.types import k8s_now
+from ..utils.fastcopy import deep_copy
 from .errors import AlreadyExistsError, ConflictError, InvalidError, NotFoundError
 from .registry import DEFAULT_REGISTRY, STATUS_SUBRESOURCE_KINDS, Registry
 
@@ -85,7 +85,9 @@ class Subscription:
         if self.namespace is not None and meta.get("namespace", "") != self.namespace:
             return
         try:
-            self._loop.call_soon_threadsafe(self._queue.put_nowait, copy.deepcopy(event))
+            # events carry a shared read-only snapshot (copied once at
+            # publish, not per subscriber) — consumers must not mutate it
+            self._loop.call_soon_threadsafe(self._queue.put_nowait, event)
         except RuntimeError:
             self._closed = True  # loop gone
 
@@ -115,6 +117,12 @@ class MemoryApiServer:
         self._rv = 0
         self._lock = threading.RLock()
         self._subs: List[Subscription] = []
+        # ownerReference uid → dependent keys (O(1) cascade GC)
+        self._by_owner: Dict[str, set] = {}
+        # Event objects are TTL'd by the real apiserver; cap them here so
+        # long-running fleets don't grow the store unboundedly
+        self._event_keys: deque = deque()
+        self.max_events = 10000
         # counters for observability/benchmarks
         self.op_counts: Dict[str, int] = {"get": 0, "list": 0, "create": 0,
                                           "update": 0, "update_status": 0, "delete": 0}
@@ -137,6 +145,22 @@ class MemoryApiServer:
         full = f"{info.plural}.{group}" if group else info.plural
         return NotFoundError(f'{full} "{name}" not found')
 
+    def _index_owners(self, key: Key, obj: Obj) -> None:
+        for ref in (obj.get("metadata") or {}).get("ownerReferences") or []:
+            uid = ref.get("uid")
+            if uid:
+                self._by_owner.setdefault(uid, set()).add(key)
+
+    def _unindex_owners(self, key: Key, obj: Obj) -> None:
+        for ref in (obj.get("metadata") or {}).get("ownerReferences") or []:
+            uid = ref.get("uid")
+            if uid:
+                deps = self._by_owner.get(uid)
+                if deps is not None:
+                    deps.discard(key)
+                    if not deps:
+                        self._by_owner.pop(uid, None)
+
     def _publish(self, ev_type: str, obj: Obj) -> None:
         event = {"type": ev_type, "object": obj}
         for sub in list(self._subs):
@@ -150,7 +174,7 @@ class MemoryApiServer:
     # -- public API --------------------------------------------------------
 
     def create(self, obj: Obj) -> Obj:
-        obj = copy.deepcopy(obj)
+        obj = deep_copy(obj)
         meta = obj.setdefault("metadata", {})
         with self._lock:
             self.op_counts["create"] += 1
@@ -176,7 +200,15 @@ class MemoryApiServer:
             meta["creationTimestamp"] = meta.get("creationTimestamp") or k8s_now()
             meta["generation"] = 1
             self._objects[key] = obj
-            out = copy.deepcopy(obj)
+            self._index_owners(key, obj)
+            if obj.get("kind") == "Event":
+                self._event_keys.append(key)
+                while len(self._event_keys) > self.max_events:
+                    old = self._event_keys.popleft()
+                    dropped = self._objects.pop(old, None)
+                    if dropped is not None:
+                        self._unindex_owners(old, dropped)
+            out = deep_copy(obj)
             self._publish("ADDED", out)
         return out
 
@@ -188,7 +220,7 @@ class MemoryApiServer:
             obj = self._objects.get((api_version, kind, ns, name))
             if obj is None:
                 raise self._not_found(api_version, kind, name)
-            return copy.deepcopy(obj)
+            return deep_copy(obj)
 
     def list(
         self,
@@ -208,11 +240,11 @@ class MemoryApiServer:
                     continue
                 if selector and not _labels_match(obj, selector):
                     continue
-                out.append(copy.deepcopy(obj))
+                out.append(deep_copy(obj))
             return out
 
     def update(self, obj: Obj) -> Obj:
-        obj = copy.deepcopy(obj)
+        obj = deep_copy(obj)
         key = self._key(obj)
         meta = obj.setdefault("metadata", {})
         with self._lock:
@@ -236,7 +268,7 @@ class MemoryApiServer:
             # status subresource: plain update cannot change status
             if (key[0], key[1]) in STATUS_SUBRESOURCE_KINDS:
                 if "status" in existing:
-                    obj["status"] = copy.deepcopy(existing["status"])
+                    obj["status"] = deep_copy(existing["status"])
                 else:
                     obj.pop("status", None)
             # no-op updates don't bump the resourceVersion or emit watch
@@ -244,25 +276,29 @@ class MemoryApiServer:
             meta["resourceVersion"] = ex_meta.get("resourceVersion")
             meta["generation"] = ex_meta.get("generation", 1)
             if obj == existing:
-                return copy.deepcopy(existing)
+                return deep_copy(existing)
             if obj.get("spec") != existing.get("spec"):
                 meta["generation"] = int(ex_meta.get("generation", 1)) + 1
             else:
                 meta["generation"] = ex_meta.get("generation", 1)
             meta["resourceVersion"] = self._next_rv()
+            self._unindex_owners(key, existing)
             self._objects[key] = obj
+            self._index_owners(key, obj)
             # finalizer removal completes a pending delete
             if meta.get("deletionTimestamp") and not meta.get("finalizers"):
                 del self._objects[key]
-                out = copy.deepcopy(obj)
+                self._unindex_owners(key, obj)
+                out = deep_copy(obj)
                 self._publish("DELETED", out)
+                self._cascade_delete(meta.get("uid"))
                 return out
-            out = copy.deepcopy(obj)
+            out = deep_copy(obj)
             self._publish("MODIFIED", out)
         return out
 
     def update_status(self, obj: Obj) -> Obj:
-        obj = copy.deepcopy(obj)
+        obj = deep_copy(obj)
         key = self._key(obj)
         meta = obj.get("metadata") or {}
         with self._lock:
@@ -278,16 +314,16 @@ class MemoryApiServer:
                     f'"{meta.get("name")}": the object has been modified; please apply '
                     f"your changes to the latest version and try again"
                 )
-            updated = copy.deepcopy(existing)
+            updated = deep_copy(existing)
             if "status" in obj:
-                updated["status"] = copy.deepcopy(obj["status"])
+                updated["status"] = deep_copy(obj["status"])
             else:
                 updated.pop("status", None)
             if updated == existing:  # no-op status write (apiserver semantics)
-                return copy.deepcopy(existing)
+                return deep_copy(existing)
             updated["metadata"]["resourceVersion"] = self._next_rv()
             self._objects[key] = updated
-            out = copy.deepcopy(updated)
+            out = deep_copy(updated)
             self._publish("MODIFIED", out)
         return out
 
@@ -305,29 +341,26 @@ class MemoryApiServer:
                 if not meta.get("deletionTimestamp"):
                     meta["deletionTimestamp"] = k8s_now()
                     meta["resourceVersion"] = self._next_rv()
-                    self._publish("MODIFIED", copy.deepcopy(obj))
+                    self._publish("MODIFIED", deep_copy(obj))
                 return
             del self._objects[key]
-            self._publish("DELETED", copy.deepcopy(obj))
+            self._unindex_owners(key, obj)
+            self._publish("DELETED", deep_copy(obj))
             self._cascade_delete(meta.get("uid"))
 
     def _cascade_delete(self, owner_uid: Optional[str]) -> None:
         """Background-propagation GC: delete dependents whose ownerReferences
         name the deleted uid (the mechanism behind Workflow cleanup on
-        HealthCheck delete, healthcheck_controller.go:512-522)."""
+        HealthCheck delete, healthcheck_controller.go:512-522). O(dependents)
+        via the owner-uid index, not a store scan."""
         if not owner_uid:
             return
-        dependents = [
-            key for key, o in self._objects.items()
-            if any(
-                ref.get("uid") == owner_uid
-                for ref in (o.get("metadata") or {}).get("ownerReferences") or []
-            )
-        ]
+        dependents = list(self._by_owner.pop(owner_uid, ()))
         for key in dependents:
             obj = self._objects.pop(key, None)
             if obj is not None:
-                self._publish("DELETED", copy.deepcopy(obj))
+                self._unindex_owners(key, obj)
+                self._publish("DELETED", deep_copy(obj))
                 self._cascade_delete(obj["metadata"].get("uid"))
 
     def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription:

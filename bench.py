@@ -1,0 +1,273 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: sustained HealthCheck reconcile throughput.
+
+Measures the metric BASELINE.json pins: p50 reconcile latency and sustained
+concurrent HealthCheck CRs on the reconcile-worker scaling axis (the workload
+is pure Kubernetes control-plane — no GPU code paths; the MI355X box serves as
+the Linux host, per BASELINE.json's north star).
+
+Per rank (one process per --gpus N slot, launched by torch.distributed.run for
+N>1): an in-memory apiserver + the full controller stack (manager, N_w
+reconcile workers, workflow watch hub, scripted workflow engine) drives a
+fleet of synthetic HealthCheck CRs shaped like BASELINE config 5 — mixed
+repeatAfterSec + cron CRs with a remedy-carrying failing fraction.
+
+One STEP = one complete wave: every CR in the fleet completes exactly one
+additional health-check run (reconcile → RBAC ensure → workflow submit →
+watch → status persisted, remedy cycle included for the failing fraction).
+W warmup waves untimed, then exactly K timed waves bracketed by a
+barrier + torch.cuda.synchronize() on both sides; rank 0 prints one JSON line
+with the whole-job aggregate (sum of per-rank cycles/s; per-rank work is fixed
+⇒ weak scaling).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import statistics
+import sys
+import time
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1, help="world size (1 rank per slot)")
+    p.add_argument("--steps", type=int, default=10, help="timed waves")
+    p.add_argument("--warmup", type=int, default=2, help="untimed warmup waves")
+    p.add_argument("--crs", type=int, default=1000, help="HealthCheck CRs per rank")
+    p.add_argument("--workers", type=int, default=8, help="MaxConcurrentReconciles per rank")
+    p.add_argument("--latency", type=float, default=0.0,
+                   help="simulated apiserver RTT per request (seconds)")
+    p.add_argument("--remedy-frac", type=float, default=0.2)
+    p.add_argument("--cron-frac", type=float, default=0.3)
+    return p.parse_args()
+
+
+INLINE_WF = """\
+apiVersion: argoproj.io/v1alpha1
+kind: Workflow
+spec:
+  entrypoint: start
+  templates:
+    - name: start
+      container:
+        image: busybox
+        command: [echo, check]
+"""
+
+
+def make_cr(i: int, ns: str, cron_frac: float, remedy_frac: float):
+    """Synthetic CR mix per BASELINE config 5: mixed cron + repeatAfterSec
+    with Remedy."""
+    n_total = 1.0
+    name = f"hc-{i:05d}"
+    is_remedy = (i % 100) < remedy_frac * 100
+    is_cron = not is_remedy and (i % 100) < (remedy_frac + cron_frac) * 100
+    spec = {
+        "level": "cluster" if i % 2 == 0 else "namespace",
+        "workflow": {
+            "generateName": f"{name}-wf-",
+            "workflowtimeout": 30,
+            "resource": {
+                "namespace": ns,
+                # share SAs across groups of CRs like a real fleet would
+                "serviceAccount": f"bench-sa-{i % 16}",
+                "source": {"inline": INLINE_WF},
+            },
+        },
+    }
+    if is_cron:
+        spec["schedule"] = {"cron": "@every 1h"}  # driven by bench waves, not timers
+    else:
+        spec["repeatAfterSec"] = 3600
+    if is_remedy:
+        spec["remedyworkflow"] = {
+            "generateName": f"{name}-remedy-wf-",
+            "workflowtimeout": 30,
+            "resource": {
+                "namespace": ns,
+                # remedy SAs are created+deleted per remedy run, so a real
+                # fleet scopes them per check to avoid teardown races
+                "serviceAccount": f"bench-remedy-sa-{i}",
+                "source": {"inline": INLINE_WF},
+            },
+        }
+    return {
+        "apiVersion": "activemonitor.keikoproj.io/v1alpha1",
+        "kind": "HealthCheck",
+        "metadata": {"name": name, "namespace": ns},
+        "spec": spec,
+    }, is_remedy
+
+
+async def run_rank(args, rank: int):
+    from active_monitor_amd.engine import Manager
+    from active_monitor_amd.kube import MemoryApiServer, MemoryClient
+    from active_monitor_amd.workflow import ScriptedWorkflowEngine
+
+    ns = "health"
+    server = MemoryApiServer()
+    client = MemoryClient(server, latency=args.latency)
+
+    def policy(wf):
+        # remedy-carrying CRs have failing checks; remedies succeed
+        name = wf["metadata"]["name"]
+        if "-remedy-wf-" in name:
+            return ("Succeeded", "")
+        if name.startswith("hc-") and _is_remedy_name(name):
+            return ("Failed", "synthetic failure")
+        return ("Succeeded", "")
+
+    remedy_names = set()
+
+    def _is_remedy_name(wfname):
+        # wfname: hc-00001-wf-xxxxx → hc-00001
+        return wfname.split("-wf-")[0] in remedy_names
+
+    engine = ScriptedWorkflowEngine(client, policy=policy)
+    await engine.start()
+    manager = Manager(client, max_workers=args.workers)
+    await manager.start()
+
+    crs = []
+    for i in range(args.crs):
+        cr, is_remedy = make_cr(i, ns, args.cron_frac, args.remedy_frac)
+        if is_remedy:
+            remedy_names.add(cr["metadata"]["name"])
+        crs.append(cr)
+        await client.create(cr)
+
+    rec = manager.reconciler
+
+    async def wave(n: int = 1):
+        """Trigger one run for every CR and wait until all complete."""
+        target = rec.completed_runs + args.crs
+        for cr in crs:
+            manager.queue.add_nowait((ns, cr["metadata"]["name"]), {"timer"})
+        while rec.completed_runs < target:
+            await asyncio.sleep(0.005)
+
+    # creation-triggered first runs count as settling, not a wave
+    while rec.completed_runs < args.crs:
+        await asyncio.sleep(0.01)
+
+    for _ in range(args.warmup):
+        await wave()
+    manager.drain_latencies()
+
+    barrier()
+    t0 = time.monotonic()
+    wave_times = []
+    for _ in range(args.steps):
+        w0 = time.monotonic()
+        await wave()
+        wave_times.append(time.monotonic() - w0)
+    barrier()
+    elapsed = time.monotonic() - t0
+
+    lat = manager.drain_latencies()
+    p50 = statistics.median(lat) * 1000 if lat else 0.0
+    p99 = (sorted(lat)[int(len(lat) * 0.99)] * 1000) if lat else 0.0
+
+    await manager.stop()
+    await engine.stop()
+    return {
+        "elapsed": elapsed,
+        "cycles": args.crs * args.steps,
+        "cycles_per_sec": args.crs * args.steps / elapsed,
+        "p50_reconcile_latency_ms": p50,
+        "p99_reconcile_latency_ms": p99,
+        "wave_times": wave_times,
+        "apiserver_requests": sum(server.op_counts.values()),
+    }
+
+
+_DIST = {"initialized": False, "torch": None}
+
+
+def barrier():
+    if _DIST["initialized"]:
+        import torch.distributed as dist
+
+        dist.barrier()
+    t = _DIST["torch"]
+    if t is not None and t.cuda.is_available():
+        t.cuda.synchronize()
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+
+    try:
+        import torch
+
+        _DIST["torch"] = torch
+    except ImportError:
+        torch = None
+
+    if world > 1:
+        import torch.distributed as dist
+
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        if backend == "nccl":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
+        dist.init_process_group(backend=backend)
+        _DIST["initialized"] = True
+
+    result = asyncio.run(run_rank(args, rank))
+
+    if world > 1:
+        import torch.distributed as dist
+
+        all_results = [None] * world
+        dist.all_gather_object(all_results, result)
+    else:
+        all_results = [result]
+
+    if rank == 0:
+        total_cps = sum(r["cycles_per_sec"] for r in all_results)
+        max_elapsed = max(r["elapsed"] for r in all_results)
+        p50 = max(r["p50_reconcile_latency_ms"] for r in all_results)
+        p99 = max(r["p99_reconcile_latency_ms"] for r in all_results)
+        out = {
+            "metric": "sustained_healthcheck_cycles_per_sec",
+            "value": round(total_cps, 2),
+            "unit": "healthcheck cycles/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(max_elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "n/a",
+            "data": "synthetic",
+            "config": {
+                "model": "healthcheck-controller",
+                "global_batch": args.crs * world,
+                "seq_len": 0,
+                "parallelism": f"shard{world}x{args.workers}w",
+                "crs_per_rank": args.crs,
+                "workers_per_rank": args.workers,
+                "max_concurrent_crs": args.crs * world,
+                "cr_mix": f"{int((1-args.cron_frac-args.remedy_frac)*100)}% repeatAfterSec, "
+                          f"{int(args.cron_frac*100)}% cron, "
+                          f"{int(args.remedy_frac*100)}% failing-with-remedy",
+                "p50_reconcile_latency_ms": round(p50, 4),
+                "p99_reconcile_latency_ms": round(p99, 4),
+                "apiserver_latency_s": args.latency,
+            },
+        }
+        print(json.dumps(out))
+    if _DIST["initialized"]:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    sys.exit(main())
