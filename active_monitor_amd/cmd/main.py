@@ -1,0 +1,132 @@
+"""Controller process entry point (reference: cmd/main.go).
+
+Flag parity with the reference (cmd/main.go:138-144):
+
+- ``--metrics-bind-address`` (default ``:8443``)
+- ``--health-probe-bind-address`` (default ``:8081``)
+- ``--leader-elect`` (default off; LeaderElectionID ``689451f8.keikoproj.io``)
+- ``--max-workers`` (default 10 — MaxConcurrentReconciles)
+
+plus the backend selection this framework adds:
+
+- ``--backend memory`` (default): self-contained in-memory apiserver with the
+  local subprocess workflow engine — run health checks standalone, no cluster.
+- ``--backend http --server URL``: a real Kubernetes apiserver (workflows are
+  executed by the cluster's Argo controller, exactly like the reference).
+"""
+from __future__ import annotations
+
+import argparse
+import asyncio
+import logging
+import signal
+import sys
+from typing import Optional, Tuple
+
+
+def parse_bind_address(addr: str) -> Optional[Tuple[str, int]]:
+    """':8443' → ('0.0.0.0', 8443); '0' or '' disables."""
+    if not addr or addr == "0":
+        return None
+    host, _, port = addr.rpartition(":")
+    return (host or "0.0.0.0", int(port))
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="active-monitor-amd",
+        description="HealthCheck/Remedy controller (clean-room active-monitor)",
+    )
+    p.add_argument("--metrics-bind-address", default=":8443",
+                   help="metrics endpoint address ('0' disables)")
+    p.add_argument("--health-probe-bind-address", default=":8081",
+                   help="healthz/readyz endpoint address ('0' disables)")
+    p.add_argument("--leader-elect", action="store_true",
+                   help="enable leader election for controller manager")
+    p.add_argument("--max-workers", type=int, default=10,
+                   help="maximum number of concurrent reconciles")
+    p.add_argument("--namespace", default=None,
+                   help="restrict watch to one namespace (default: all)")
+    p.add_argument("--backend", choices=["memory", "http"], default="memory")
+    p.add_argument("--server", default="",
+                   help="apiserver URL for --backend http")
+    p.add_argument("--token", default="", help="bearer token for --backend http")
+    p.add_argument("--insecure-skip-tls-verify", action="store_true")
+    p.add_argument("--workflow-engine", choices=["local", "none"], default="local",
+                   help="memory backend: execute workflows locally or leave to an "
+                        "external engine")
+    p.add_argument("--zap-log-level", default="info",
+                   help="log level (debug/info/warn/error)")
+    return p
+
+
+async def run(args) -> int:
+    from ..engine import Manager
+    from ..kube import MemoryApiServer, MemoryClient
+
+    level = {"debug": logging.DEBUG, "info": logging.INFO,
+             "warn": logging.WARNING, "error": logging.ERROR}.get(
+        args.zap_log_level, logging.INFO)
+    logging.basicConfig(
+        level=level,
+        format="%(asctime)s %(levelname)s %(name)s %(message)s",
+    )
+    log = logging.getLogger("active_monitor_amd.main")
+
+    engine = None
+    if args.backend == "http":
+        from ..kube.http import HttpClient
+
+        client = HttpClient(
+            args.server, token=args.token or None,
+            verify=not args.insecure_skip_tls_verify,
+        )
+        await client.start()
+    else:
+        client = MemoryClient(MemoryApiServer())
+        if args.workflow_engine == "local":
+            from ..workflow import LocalWorkflowEngine
+
+            engine = LocalWorkflowEngine(client, args.namespace)
+
+    manager = Manager(
+        client,
+        max_workers=args.max_workers,
+        namespace=args.namespace,
+        metrics_addr=parse_bind_address(args.metrics_bind_address),
+        health_addr=parse_bind_address(args.health_probe_bind_address),
+        leader_elect=args.leader_elect,
+    )
+
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    for sig in (signal.SIGINT, signal.SIGTERM):
+        try:
+            loop.add_signal_handler(sig, stop.set)
+        except NotImplementedError:  # pragma: no cover - non-unix
+            pass
+
+    if engine is not None:
+        await engine.start()
+    await manager.start()
+    log.info("starting manager: workers=%d backend=%s", args.max_workers, args.backend)
+    await stop.wait()
+    log.info("shutting down")
+    await manager.stop()
+    if engine is not None:
+        await engine.stop()
+    if args.backend == "http":
+        await client.close()
+    return 0
+
+
+def main(argv=None) -> int:
+    args = build_parser().parse_args(argv)
+    try:
+        return asyncio.run(run(args))
+    except KeyboardInterrupt:
+        return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

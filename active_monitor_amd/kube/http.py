@@ -1,0 +1,226 @@
+"""HTTP Kubernetes client — the real-apiserver backend.
+
+Implements the same async KubeClient facade the reconciler programs against
+(kube/client.py), over the Kubernetes REST API: typed/namespaced paths,
+status subresource writes, label selectors, and streaming watches with
+reconnect — the role client-go's typed + dynamic clients play in the
+reference (healthcheck_controller.go:133-137).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+import ssl
+from typing import Any, AsyncIterator, Dict, List, Optional
+
+import aiohttp
+
+from .errors import (
+    AlreadyExistsError,
+    ApiError,
+    ConflictError,
+    InvalidError,
+    NotFoundError,
+)
+from .registry import DEFAULT_REGISTRY, Registry
+
+log = logging.getLogger("active_monitor_amd.kube.http")
+
+Obj = Dict[str, Any]
+
+
+def _error_for(status: int, body: str) -> ApiError:
+    reason = ""
+    message = body
+    try:
+        parsed = json.loads(body)
+        reason = parsed.get("reason", "")
+        message = parsed.get("message", body)
+    except (ValueError, TypeError):
+        pass
+    if status == 404:
+        return NotFoundError(message)
+    if status == 409:
+        if reason == "AlreadyExists":
+            return AlreadyExistsError(message)
+        return ConflictError(message)
+    if status == 422:
+        return InvalidError(message)
+    err = ApiError(message)
+    err.code = status
+    return err
+
+
+class HttpSubscription:
+    """Streaming watch with automatic reconnect (resourceVersion resume)."""
+
+    def __init__(self, client: "HttpClient", api_version: str, kind: str,
+                 namespace: Optional[str]):
+        self._client = client
+        self.api_version = api_version
+        self.kind = kind
+        self.namespace = namespace
+        self._closed = False
+        self._resource_version: Optional[str] = None
+        self._queue: "asyncio.Queue[Optional[Dict[str, Any]]]" = asyncio.Queue()
+        self._task = asyncio.ensure_future(self._pump())
+
+    async def _pump(self) -> None:
+        while not self._closed:
+            try:
+                await self._stream_once()
+            except asyncio.CancelledError:
+                return
+            except Exception as e:
+                if self._closed:
+                    return
+                log.warning("watch stream error (%s/%s): %s; reconnecting",
+                            self.api_version, self.kind, e)
+                await asyncio.sleep(1.0)
+
+    async def _stream_once(self) -> None:
+        params = {"watch": "true"}
+        if self._resource_version:
+            params["resourceVersion"] = self._resource_version
+        path = self._client._collection_path(self.api_version, self.kind, self.namespace)
+        async with self._client._session.get(
+            self._client.base_url + path, params=params,
+            timeout=aiohttp.ClientTimeout(total=None, sock_read=None),
+        ) as resp:
+            if resp.status >= 400:
+                raise _error_for(resp.status, await resp.text())
+            async for line in resp.content:
+                if self._closed:
+                    return
+                line = line.strip()
+                if not line:
+                    continue
+                ev = json.loads(line)
+                obj = ev.get("object") or {}
+                rv = (obj.get("metadata") or {}).get("resourceVersion")
+                if rv:
+                    self._resource_version = str(rv)
+                if ev.get("type") in ("ADDED", "MODIFIED", "DELETED"):
+                    self._queue.put_nowait({"type": ev["type"], "object": obj})
+
+    def close(self) -> None:
+        self._closed = True
+        self._task.cancel()
+        self._queue.put_nowait(None)
+
+    def __aiter__(self) -> AsyncIterator[Dict[str, Any]]:
+        return self
+
+    async def __anext__(self) -> Dict[str, Any]:
+        ev = await self._queue.get()
+        if ev is None:
+            raise StopAsyncIteration
+        return ev
+
+
+class HttpClient:
+    def __init__(
+        self,
+        base_url: str,
+        token: Optional[str] = None,
+        verify: bool = True,
+        ca_cert: Optional[str] = None,
+        registry: Registry = DEFAULT_REGISTRY,
+    ):
+        self.base_url = base_url.rstrip("/")
+        self.token = token
+        self.verify = verify
+        self.ca_cert = ca_cert
+        self.registry = registry
+        self._session: Optional[aiohttp.ClientSession] = None
+
+    async def start(self) -> None:
+        headers = {"Content-Type": "application/json"}
+        if self.token:
+            headers["Authorization"] = f"Bearer {self.token}"
+        ssl_ctx: Any = None
+        if self.base_url.startswith("https"):
+            if not self.verify:
+                ssl_ctx = ssl.create_default_context()
+                ssl_ctx.check_hostname = False
+                ssl_ctx.verify_mode = ssl.CERT_NONE
+            elif self.ca_cert:
+                ssl_ctx = ssl.create_default_context(cafile=self.ca_cert)
+        self._session = aiohttp.ClientSession(
+            headers=headers, connector=aiohttp.TCPConnector(ssl=ssl_ctx)
+        )
+
+    async def close(self) -> None:
+        if self._session is not None:
+            await self._session.close()
+
+    # -- paths --------------------------------------------------------------
+
+    def _collection_path(self, api_version: str, kind: str, namespace: Optional[str]) -> str:
+        info = self.registry.by_kind(api_version, kind)
+        prefix = f"/api/{api_version}" if "/" not in api_version else f"/apis/{api_version}"
+        if info.namespaced and namespace:
+            return f"{prefix}/namespaces/{namespace}/{info.plural}"
+        return f"{prefix}/{info.plural}"
+
+    def _object_path(self, api_version: str, kind: str, namespace: str, name: str) -> str:
+        return f"{self._collection_path(api_version, kind, namespace or None)}/{name}"
+
+    # -- requests -----------------------------------------------------------
+
+    async def _request(self, method: str, path: str, body: Optional[Obj] = None,
+                       params: Optional[Dict[str, str]] = None) -> Obj:
+        assert self._session is not None, "call start() first"
+        async with self._session.request(
+            method, self.base_url + path,
+            json=body if body is not None else None, params=params,
+        ) as resp:
+            text = await resp.text()
+            if resp.status >= 400:
+                raise _error_for(resp.status, text)
+            return json.loads(text) if text else {}
+
+    async def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+        return await self._request("GET", self._object_path(api_version, kind, namespace, name))
+
+    async def list(
+        self, api_version: str, kind: str,
+        namespace: Optional[str] = None, label_selector: Optional[str] = None,
+    ) -> List[Obj]:
+        params = {}
+        if label_selector:
+            params["labelSelector"] = label_selector
+        out = await self._request(
+            "GET", self._collection_path(api_version, kind, namespace), params=params
+        )
+        return out.get("items", [])
+
+    async def create(self, obj: Obj) -> Obj:
+        meta = obj.get("metadata") or {}
+        path = self._collection_path(
+            obj.get("apiVersion", ""), obj.get("kind", ""), meta.get("namespace")
+        )
+        return await self._request("POST", path, obj)
+
+    async def update(self, obj: Obj) -> Obj:
+        meta = obj.get("metadata") or {}
+        path = self._object_path(
+            obj.get("apiVersion", ""), obj.get("kind", ""),
+            meta.get("namespace", ""), meta.get("name", ""),
+        )
+        return await self._request("PUT", path, obj)
+
+    async def update_status(self, obj: Obj) -> Obj:
+        meta = obj.get("metadata") or {}
+        path = self._object_path(
+            obj.get("apiVersion", ""), obj.get("kind", ""),
+            meta.get("namespace", ""), meta.get("name", ""),
+        ) + "/status"
+        return await self._request("PUT", path, obj)
+
+    async def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None:
+        await self._request("DELETE", self._object_path(api_version, kind, namespace, name))
+
+    def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> HttpSubscription:
+        return HttpSubscription(self, api_version, kind, namespace)

@@ -1,0 +1,157 @@
+"""HTTP apiserver frontend: serves the in-memory store over the Kubernetes
+REST API.
+
+This is the framework's envtest-style local apiserver: the HttpClient (and
+kubectl-shaped tooling) can talk to a MemoryApiServer over real HTTP —
+typed paths, status subresource, label selectors, and streaming watches
+(JSON-lines, resourceVersion resume semantics are best-effort: a reconnect
+replays current state as ADDED events, which the informer path tolerates).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import logging
+from typing import Optional
+
+from aiohttp import web
+
+from .errors import ApiError
+from .memory import MemoryApiServer
+
+log = logging.getLogger("active_monitor_amd.kube.server")
+
+
+def _status_body(err: ApiError) -> dict:
+    return {
+        "kind": "Status",
+        "apiVersion": "v1",
+        "status": "Failure",
+        "message": err.message,
+        "reason": err.reason,
+        "code": err.code,
+    }
+
+
+class ApiServerFrontend:
+    def __init__(self, server: MemoryApiServer, host: str = "127.0.0.1", port: int = 0):
+        self.server = server
+        self.host = host
+        self.port = port
+        self._runner: Optional[web.AppRunner] = None
+
+    @property
+    def url(self) -> str:
+        return f"http://{self.host}:{self.port}"
+
+    async def start(self) -> None:
+        app = web.Application()
+        app.router.add_route("*", "/api/{version}/{tail:.*}", self._handle_core)
+        app.router.add_route("*", "/apis/{group}/{version}/{tail:.*}", self._handle_group)
+        self._runner = web.AppRunner(app, access_log=None)
+        await self._runner.setup()
+        site = web.TCPSite(self._runner, self.host, self.port)
+        await site.start()
+        if self.port == 0:
+            self.port = site._server.sockets[0].getsockname()[1]
+
+    async def stop(self) -> None:
+        if self._runner is not None:
+            await self._runner.cleanup()
+
+    # -- routing -----------------------------------------------------------
+
+    async def _handle_core(self, request: web.Request) -> web.StreamResponse:
+        version = request.match_info["version"]
+        return await self._dispatch(request, version, request.match_info["tail"])
+
+    async def _handle_group(self, request: web.Request) -> web.StreamResponse:
+        api_version = f'{request.match_info["group"]}/{request.match_info["version"]}'
+        return await self._dispatch(request, api_version, request.match_info["tail"])
+
+    async def _dispatch(self, request: web.Request, api_version: str, tail: str) -> web.StreamResponse:
+        # tail forms:
+        #   {plural}
+        #   {plural}/{name}
+        #   {plural}/{name}/status
+        #   namespaces/{ns}/{plural}
+        #   namespaces/{ns}/{plural}/{name}
+        #   namespaces/{ns}/{plural}/{name}/status
+        parts = [p for p in tail.split("/") if p]
+        namespace = ""
+        if len(parts) >= 2 and parts[0] == "namespaces" and (
+            len(parts) > 2 or request.method in ("GET", "DELETE")
+        ):
+            # disambiguate /api/v1/namespaces/{name} (the Namespace resource)
+            # from /api/v1/namespaces/{ns}/{plural}...
+            if len(parts) == 2:
+                parts = ["namespaces", parts[1]]  # the Namespace object itself
+            else:
+                namespace = parts[1]
+                parts = parts[2:]
+        plural = parts[0] if parts else ""
+        name = parts[1] if len(parts) > 1 else ""
+        subresource = parts[2] if len(parts) > 2 else ""
+
+        try:
+            info = self.server.registry.by_plural(api_version, plural)
+        except KeyError:
+            return web.json_response(
+                _status_body(ApiError(f"unknown resource {plural}")), status=404
+            )
+        kind = info.kind
+
+        try:
+            if request.method == "GET" and not name:
+                if request.query.get("watch") in ("true", "1"):
+                    return await self._watch(request, api_version, kind, namespace or None)
+                items = self.server.list(
+                    api_version, kind, namespace or None,
+                    request.query.get("labelSelector"),
+                )
+                return web.json_response(
+                    {"apiVersion": api_version, "kind": kind + "List", "items": items}
+                )
+            if request.method == "GET":
+                return web.json_response(self.server.get(api_version, kind, namespace, name))
+            if request.method == "POST":
+                obj = await request.json()
+                meta = obj.setdefault("metadata", {})
+                if namespace and not meta.get("namespace"):
+                    meta["namespace"] = namespace
+                return web.json_response(self.server.create(obj), status=201)
+            if request.method == "PUT" and subresource == "status":
+                return web.json_response(self.server.update_status(await request.json()))
+            if request.method == "PUT":
+                return web.json_response(self.server.update(await request.json()))
+            if request.method == "DELETE":
+                self.server.delete(api_version, kind, namespace, name)
+                return web.json_response({"kind": "Status", "status": "Success"})
+        except ApiError as e:
+            return web.json_response(_status_body(e), status=e.code)
+        return web.json_response(
+            _status_body(ApiError(f"unsupported method {request.method}")), status=405
+        )
+
+    async def _watch(self, request: web.Request, api_version: str, kind: str,
+                     namespace: Optional[str]) -> web.StreamResponse:
+        resp = web.StreamResponse(
+            status=200, headers={"Content-Type": "application/json;stream=watch"}
+        )
+        await resp.prepare(request)
+        sub = self.server.watch(api_version, kind, namespace)
+        try:
+            # replay current state as ADDED unless resuming from an rv — the
+            # standard list-then-watch contract
+            if not request.query.get("resourceVersion"):
+                for obj in self.server.list(api_version, kind, namespace):
+                    await resp.write(
+                        (json.dumps({"type": "ADDED", "object": obj}) + "\n").encode()
+                    )
+            async for ev in sub:
+                await resp.write((json.dumps(ev) + "\n").encode())
+        except (ConnectionResetError, asyncio.CancelledError):
+            pass
+        finally:
+            sub.close()
+        return resp
