@@ -187,3 +187,6 @@ class WorkQueue:
 
     def __len__(self) -> int:
         return len(self._queue)
+
+    def __bool__(self) -> bool:
+        return True  # an empty queue is still a queue

@@ -46,7 +46,8 @@ class MemoryClient:
     """
 
     def __init__(self, server: Optional[MemoryApiServer] = None, latency: float = 0.0):
-        self.server = server or MemoryApiServer()
+        # NB: not `server or ...` — an empty MemoryApiServer is falsy (__len__)
+        self.server = server if server is not None else MemoryApiServer()
         self.latency = latency
 
     async def _lat(self) -> None:

@@ -373,3 +373,6 @@ class MemoryApiServer:
     def __len__(self) -> int:
         with self._lock:
             return len(self._objects)
+
+    def __bool__(self) -> bool:
+        return True  # an empty store is still a store (see MemoryClient note)

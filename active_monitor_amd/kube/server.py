@@ -39,6 +39,7 @@ class ApiServerFrontend:
         self.host = host
         self.port = port
         self._runner: Optional[web.AppRunner] = None
+        self._live_subs: list = []
 
     @property
     def url(self) -> str:
@@ -48,7 +49,7 @@ class ApiServerFrontend:
         app = web.Application()
         app.router.add_route("*", "/api/{version}/{tail:.*}", self._handle_core)
         app.router.add_route("*", "/apis/{group}/{version}/{tail:.*}", self._handle_group)
-        self._runner = web.AppRunner(app, access_log=None)
+        self._runner = web.AppRunner(app, access_log=None, shutdown_timeout=1.0)
         await self._runner.setup()
         site = web.TCPSite(self._runner, self.host, self.port)
         await site.start()
@@ -56,6 +57,8 @@ class ApiServerFrontend:
             self.port = site._server.sockets[0].getsockname()[1]
 
     async def stop(self) -> None:
+        for sub in list(self._live_subs):
+            sub.close()  # end open watch streams so cleanup() is immediate
         if self._runner is not None:
             await self._runner.cleanup()
 
@@ -140,6 +143,7 @@ class ApiServerFrontend:
         )
         await resp.prepare(request)
         sub = self.server.watch(api_version, kind, namespace)
+        self._live_subs.append(sub)
         try:
             # replay current state as ADDED unless resuming from an rv — the
             # standard list-then-watch contract
@@ -154,4 +158,6 @@ class ApiServerFrontend:
             pass
         finally:
             sub.close()
+            if sub in self._live_subs:
+                self._live_subs.remove(sub)
         return resp
