@@ -1,0 +1,26 @@
+# Developer entry points (parity with the reference Makefile's test/build/run
+# targets, adapted to the Python stack).
+PY ?= python3
+
+.PHONY: test test-gpu bench run manifests lint fmt docker-build sweep
+
+test:
+	$(PY) -m pytest tests/ -q -m "not gpu"
+
+test-gpu:
+	$(PY) -m pytest tests/ -q -m gpu
+
+bench:
+	$(PY) bench.py --steps 10 --warmup 2
+
+sweep:
+	$(PY) benchmarks/worker_sweep.py
+
+run:
+	$(PY) -m active_monitor_amd.cmd.main --backend memory --max-workers 10
+
+manifests:
+	$(PY) -m active_monitor_amd.api.crd > config/crd/bases/activemonitor.keikoproj.io_healthchecks.yaml
+
+docker-build:
+	docker build -t active-monitor-amd:latest .
