@@ -131,17 +131,17 @@ def test_mini_fleet_soak(run):
                 return rec.completed_runs >= 150
 
             await env.wait_for(settled, timeout=30, msg="first wave")
-            target = rec.completed_runs + 150
             for i in range(150):
                 env.manager.queue.add_nowait(("health", f"fleet-{i:03d}"), {"timer"})
 
-            async def second():
-                return rec.completed_runs >= target
+            async def all_advanced():
+                objs = await env.client.list(API_VERSION, "HealthCheck", "health")
+                return all(
+                    (o.get("status") or {}).get("totalHealthCheckRuns", 0) >= 2
+                    for o in objs
+                )
 
-            await env.wait_for(second, timeout=30, msg="second wave")
-            # every CR advanced
-            for i in range(0, 150, 37):
-                hc = await env.get_hc(f"fleet-{i:03d}")
-                assert hc.status.total_healthcheck_runs >= 2
+            await env.wait_for(all_advanced, timeout=30, interval=0.2,
+                               msg="second wave on every CR")
 
     run(go(), timeout=90)
