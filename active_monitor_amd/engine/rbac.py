@@ -20,7 +20,7 @@ from typing import Any, Dict, List, Optional
 
 from ..api.types import HealthCheck, PolicyRule
 from ..kube.client import EventRecorder, KubeClient
-from ..kube.errors import NotFoundError
+from ..kube.errors import AlreadyExistsError, NotFoundError
 
 RBAC_API_VERSION = "rbac.authorization.k8s.io/v1"
 
@@ -112,7 +112,12 @@ class RBACProvisioner:
             "kind": "ServiceAccount",
             "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
         }
-        created = await self.client.create(sa)
+        try:
+            created = await self.client.create(sa)
+        except AlreadyExistsError:
+            # concurrent reconciles sharing an SA race get-then-create;
+            # losing the race means the object exists — reuse it
+            return name
         return created["metadata"]["name"]
 
     async def create_cluster_role(self, name: str, rules: List[PolicyRule]) -> str:
@@ -127,7 +132,12 @@ class RBACProvisioner:
             "metadata": {"name": name, "labels": _managed_labels()},
             "rules": [r.to_dict() for r in rules],
         }
-        created = await self.client.create(cr)
+        try:
+            created = await self.client.create(cr)
+        except AlreadyExistsError:
+            # concurrent reconciles sharing an SA race get-then-create;
+            # losing the race means the object exists — reuse it
+            return name
         return created["metadata"]["name"]
 
     async def create_cluster_role_binding(
@@ -151,7 +161,12 @@ class RBACProvisioner:
                 {"kind": "ServiceAccount", "name": sa_name, "namespace": sa_namespace}
             ],
         }
-        created = await self.client.create(crb)
+        try:
+            created = await self.client.create(crb)
+        except AlreadyExistsError:
+            # concurrent reconciles sharing an SA race get-then-create;
+            # losing the race means the object exists — reuse it
+            return name
         return created["metadata"]["name"]
 
     async def create_namespace_role(
@@ -168,7 +183,12 @@ class RBACProvisioner:
             "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
             "rules": [r.to_dict() for r in rules],
         }
-        created = await self.client.create(role)
+        try:
+            created = await self.client.create(role)
+        except AlreadyExistsError:
+            # concurrent reconciles sharing an SA race get-then-create;
+            # losing the race means the object exists — reuse it
+            return name
         return created["metadata"]["name"]
 
     async def create_namespace_role_binding(
@@ -192,7 +212,12 @@ class RBACProvisioner:
                 {"kind": "ServiceAccount", "name": sa_name, "namespace": namespace}
             ],
         }
-        created = await self.client.create(rb)
+        try:
+            created = await self.client.create(rb)
+        except AlreadyExistsError:
+            # concurrent reconciles sharing an SA race get-then-create;
+            # losing the race means the object exists — reuse it
+            return name
         return created["metadata"]["name"]
 
     # -- deletes (only objects carrying the managed-by label) --------------

@@ -267,6 +267,16 @@ class HealthCheckReconciler:
             except CronParseError as e:
                 await self._event(hc, "Warning", "Fail to parse cron")
                 raise e
+            # the reference's elif chain exempts cron CRs from the
+            # already-scheduled dedup, so every status write re-submits them
+            # in a tight loop until the blocking watch throttles it; apply the
+            # dedup here with the freshly computed interval instead
+            if (
+                not from_timer
+                and int(time.time() - finished_unix) < hc.spec.repeat_after_sec
+                and self.get_timer_by_name(hc.name) is not None
+            ):
+                return ReconcileResult()
         elif (
             not from_timer
             and int(time.time() - finished_unix) < spec.repeat_after_sec
@@ -519,6 +529,17 @@ class HealthCheckReconciler:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):
             return
+        # cron CRs: recompute the delay at completion time so the next run
+        # lands on the cron tick, not submit-time + interval (the reference
+        # re-arms with the stale submit-time value, drifting by the run's
+        # duration — :746-752)
+        fresh_spec = fresh.get("spec") or {}
+        cron = ((fresh_spec.get("schedule") or {}).get("cron")) or ""
+        if int(fresh_spec.get("repeatAfterSec", 0) or 0) <= 0 and cron:
+            try:
+                repeat_after_sec = seconds_until_next(cron)
+            except CronParseError:
+                pass  # keep the submit-time value
         self._arm_repeat_timer(hc.name, hc.namespace, repeat_after_sec)
         try:
             await self.update_healthcheck_status(hc)
