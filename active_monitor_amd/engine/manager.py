@@ -111,6 +111,9 @@ class Manager:
             self._sub.close()
         if getattr(self, "wf_hub", None) is not None:
             await self.wf_hub.stop()
+        stop_rec = getattr(self.recorder, "stop", None)
+        if stop_rec is not None:
+            await stop_rec()
         await self.queue.shutdown()
         self.reconciler.stop_all()
         for t in self._tasks:

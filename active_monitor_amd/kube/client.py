@@ -91,14 +91,40 @@ class MemoryClient:
 class EventRecorder:
     """Kubernetes Event emission (record.EventRecorder equivalent — the
     reference emits events on nearly every transition, e.g.
-    healthcheck_controller.go:243,280,532,636,663)."""
+    healthcheck_controller.go:243,280,532,636,663).
+
+    Like client-go's event broadcaster, emission is non-blocking: ``event()``
+    enqueues and returns; a background task writes to the apiserver. Events
+    are best-effort and never slow the reconcile hot path."""
 
     NORMAL = "Normal"
     WARNING = "Warning"
 
-    def __init__(self, client: KubeClient, component: str = "active-monitor"):
+    def __init__(self, client: KubeClient, component: str = "active-monitor",
+                 buffer: int = 4096):
         self.client = client
         self.component = component
+        self._queue: Optional["asyncio.Queue[Obj]"] = None
+        self._task: Optional[asyncio.Task] = None
+        self._buffer = buffer
+        self.dropped = 0
+
+    def _ensure_pump(self) -> None:
+        if self._task is None or self._task.done():
+            self._queue = asyncio.Queue(maxsize=self._buffer)
+            self._task = asyncio.get_running_loop().create_task(self._pump())
+
+    async def _pump(self) -> None:
+        while True:
+            ev = await self._queue.get()
+            try:
+                await self.client.create(ev)
+            except asyncio.CancelledError:
+                raise
+            except Exception:  # best-effort
+                pass
+            finally:
+                self._queue.task_done()
 
     async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
         meta = involved.get("metadata") or {}
@@ -120,10 +146,27 @@ class EventRecorder:
             "source": {"component": self.component},
             "firstTimestamp": None,
         }
+        self._ensure_pump()
         try:
-            await self.client.create(ev)
-        except Exception:  # events are best-effort, never fail the caller
-            pass
+            self._queue.put_nowait(ev)
+        except asyncio.QueueFull:
+            self.dropped += 1  # drop rather than block (broadcaster behavior)
+
+    async def flush(self, timeout: float = 5.0) -> None:
+        if self._queue is not None:
+            try:
+                await asyncio.wait_for(self._queue.join(), timeout)
+            except asyncio.TimeoutError:
+                pass
+
+    async def stop(self) -> None:
+        await self.flush(1.0)
+        if self._task is not None:
+            self._task.cancel()
+            try:
+                await self._task
+            except (asyncio.CancelledError, Exception):
+                pass
 
 
 class FakeRecorder(EventRecorder):
@@ -134,6 +177,8 @@ class FakeRecorder(EventRecorder):
     def __init__(self, capacity: int = 100):
         self.events: List[str] = []
         self.capacity = capacity
+        self._queue = None
+        self._task = None
 
     async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
         if len(self.events) < self.capacity:
