@@ -27,9 +27,19 @@ from .. import API_VERSION
 # ---------------------------------------------------------------------------
 
 
+_NOW_CACHE = [0, ""]
+
+
 def k8s_now() -> str:
-    """Current UTC time in Kubernetes metav1.Time wire format."""
-    return format_k8s_time(datetime.now(timezone.utc))
+    """Current UTC time in Kubernetes metav1.Time wire format (second
+    precision, so the formatted string is cached per second)."""
+    import time as _time
+
+    sec = int(_time.time())
+    if _NOW_CACHE[0] != sec:
+        _NOW_CACHE[0] = sec
+        _NOW_CACHE[1] = format_k8s_time(datetime.fromtimestamp(sec, timezone.utc))
+    return _NOW_CACHE[1]
 
 
 def format_k8s_time(dt: datetime) -> str:

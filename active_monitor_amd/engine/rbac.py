@@ -113,7 +113,7 @@ class RBACProvisioner:
             "metadata": {"name": name, "namespace": namespace, "labels": _managed_labels()},
         }
         try:
-            created = await self.client.create(sa)
+            created = await self.client.create(sa, transfer=True)
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
@@ -133,7 +133,7 @@ class RBACProvisioner:
             "rules": [r.to_dict() for r in rules],
         }
         try:
-            created = await self.client.create(cr)
+            created = await self.client.create(cr, transfer=True)
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
@@ -162,7 +162,7 @@ class RBACProvisioner:
             ],
         }
         try:
-            created = await self.client.create(crb)
+            created = await self.client.create(crb, transfer=True)
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
@@ -184,7 +184,7 @@ class RBACProvisioner:
             "rules": [r.to_dict() for r in rules],
         }
         try:
-            created = await self.client.create(role)
+            created = await self.client.create(role, transfer=True)
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
@@ -213,7 +213,7 @@ class RBACProvisioner:
             ],
         }
         try:
-            created = await self.client.create(rb)
+            created = await self.client.create(rb, transfer=True)
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it

@@ -192,7 +192,8 @@ class HealthCheckReconciler:
         flags = flags or set()
         self.reconcile_count += 1
         try:
-            obj = await self.client.get(API_VERSION, HC_KIND, namespace, name)
+            obj = await self.client.get(API_VERSION, HC_KIND, namespace, name,
+                                        snapshot_read=True)
         except NotFoundError:
             # CR deleted: stop the repeat timer so self-scheduling halts
             # (:180-184); in-flight watches are cancelled proactively (the
@@ -328,7 +329,7 @@ class HealthCheckReconciler:
             },
             "spec": spec,
         }
-        created = await self.client.create(wf)
+        created = await self.client.create(wf, transfer=True)
         await self._event(hc, "Normal", "Successfully created workflow")
         return created["metadata"]["name"]
 
@@ -351,7 +352,7 @@ class HealthCheckReconciler:
             },
             "spec": spec,
         }
-        created = await self.client.create(wf)
+        created = await self.client.create(wf, transfer=True)
         await self._event(hc, "Normal", "Successfully created remedyWorkflow")
         return created["metadata"]["name"]
 
@@ -360,7 +361,8 @@ class HealthCheckReconciler:
     # ------------------------------------------------------------------
 
     async def _poll_workflow(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
-        wf = await self.client.get(WF_API_VERSION, WF_KIND, namespace, name)
+        wf = await self.client.get(WF_API_VERSION, WF_KIND, namespace, name,
+                                   snapshot_read=True)
         status = wf.get("status")
         return status if isinstance(status, dict) else None
 
@@ -524,7 +526,8 @@ class HealthCheckReconciler:
         closing a duplicate-submission race the reference leaves open by
         updating first (:734) and arming after (:746)."""
         try:
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
+                                          snapshot_read=True)
         except NotFoundError:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):
@@ -645,7 +648,8 @@ class HealthCheckReconciler:
 
         # persist remedy status promptly (reference :856-871)
         try:
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
+                                          snapshot_read=True)
         except NotFoundError:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):
@@ -660,7 +664,8 @@ class HealthCheckReconciler:
         """Fresh Get + status-subresource update with conflict retry."""
         last: Optional[BaseException] = None
         for attempt in range(retries):
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name)
+            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
+                                          snapshot_read=True)
             fresh["status"] = hc.status.to_dict()
             try:
                 await self.client.update_status(fresh)

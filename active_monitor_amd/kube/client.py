@@ -50,27 +50,34 @@ class MemoryClient:
         self.server = server if server is not None else MemoryApiServer()
         self.latency = latency
 
+    _yield_counter = 0
+
     async def _lat(self) -> None:
         if self.latency > 0:
             await asyncio.sleep(self.latency)
         else:
-            # cooperative yield so tight loops don't starve the event loop
-            await asyncio.sleep(0)
+            # amortized cooperative yield: enough to keep tight loops from
+            # starving the event loop without paying a call_soon per request
+            MemoryClient._yield_counter += 1
+            if MemoryClient._yield_counter % 16 == 0:
+                await asyncio.sleep(0)
 
-    async def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+    async def get(self, api_version: str, kind: str, namespace: str, name: str,
+                  snapshot_read: bool = False) -> Obj:
         await self._lat()
-        return self.server.get(api_version, kind, namespace, name)
+        return self.server.get(api_version, kind, namespace, name, snapshot_read)
 
     async def list(
         self, api_version: str, kind: str,
         namespace: Optional[str] = None, label_selector: Optional[str] = None,
+        snapshot_read: bool = False,
     ) -> List[Obj]:
         await self._lat()
-        return self.server.list(api_version, kind, namespace, label_selector)
+        return self.server.list(api_version, kind, namespace, label_selector, snapshot_read)
 
-    async def create(self, obj: Obj) -> Obj:
+    async def create(self, obj: Obj, transfer: bool = False) -> Obj:
         await self._lat()
-        return self.server.create(obj)
+        return self.server.create(obj, transfer)
 
     async def update(self, obj: Obj) -> Obj:
         await self._lat()
@@ -118,7 +125,11 @@ class EventRecorder:
         while True:
             ev = await self._queue.get()
             try:
-                await self.client.create(ev)
+                create = self.client.create
+                try:
+                    await create(ev, transfer=True)
+                except TypeError:  # backend without transfer support
+                    await create(ev)
             except asyncio.CancelledError:
                 raise
             except Exception:  # best-effort

@@ -181,12 +181,16 @@ class HttpClient:
                 raise _error_for(resp.status, text)
             return json.loads(text) if text else {}
 
-    async def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+    async def get(self, api_version: str, kind: str, namespace: str, name: str,
+                  snapshot_read: bool = False) -> Obj:
+        # snapshot_read is a memory-backend optimization; wire responses are
+        # always private copies
         return await self._request("GET", self._object_path(api_version, kind, namespace, name))
 
     async def list(
         self, api_version: str, kind: str,
         namespace: Optional[str] = None, label_selector: Optional[str] = None,
+        snapshot_read: bool = False,
     ) -> List[Obj]:
         params = {}
         if label_selector:
@@ -196,7 +200,8 @@ class HttpClient:
         )
         return out.get("items", [])
 
-    async def create(self, obj: Obj) -> Obj:
+    async def create(self, obj: Obj, transfer: bool = False) -> Obj:
+        # transfer is a memory-backend optimization; ignored on the wire
         meta = obj.get("metadata") or {}
         path = self._collection_path(
             obj.get("apiVersion", ""), obj.get("kind", ""), meta.get("namespace")

@@ -27,7 +27,7 @@ from collections import deque
 from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
 
 from ..api.types import k8s_now
-from ..utils.fastcopy import deep_copy
+from ..utils.fastcopy import deep_copy, snapshot
 from .errors import AlreadyExistsError, ConflictError, InvalidError, NotFoundError
 from .registry import DEFAULT_REGISTRY, STATUS_SUBRESOURCE_KINDS, Registry
 
@@ -38,7 +38,7 @@ _SUFFIX_ALPHABET = "bcdfghjklmnpqrstvwxz2456789"  # k8s-style name suffix chars
 
 
 def _rand_suffix(n: int = 5) -> str:
-    return "".join(random.choice(_SUFFIX_ALPHABET) for _ in range(n))
+    return "".join(random.choices(_SUFFIX_ALPHABET, k=n))
 
 
 def parse_label_selector(selector: Optional[str]) -> Dict[str, str]:
@@ -173,8 +173,13 @@ class MemoryApiServer:
 
     # -- public API --------------------------------------------------------
 
-    def create(self, obj: Obj) -> Obj:
-        obj = deep_copy(obj)
+    def create(self, obj: Obj, transfer: bool = False) -> Obj:
+        """``transfer=True`` lets the store take ownership of ``obj`` (no
+        copy-in) and return a read-optimized snapshot (no copy-out). Only for
+        callers that built the dict themselves and never touch it again —
+        the controller's submit/event/RBAC paths."""
+        if not transfer:
+            obj = deep_copy(obj)
         meta = obj.setdefault("metadata", {})
         with self._lock:
             self.op_counts["create"] += 1
@@ -208,11 +213,18 @@ class MemoryApiServer:
                     dropped = self._objects.pop(old, None)
                     if dropped is not None:
                         self._unindex_owners(old, dropped)
-            out = deep_copy(obj)
+            out = snapshot(obj)
             self._publish("ADDED", out)
+            if not transfer:
+                out = deep_copy(obj)
         return out
 
-    def get(self, api_version: str, kind: str, namespace: str, name: str) -> Obj:
+    def get(self, api_version: str, kind: str, namespace: str, name: str,
+            snapshot_read: bool = False) -> Obj:
+        """``snapshot_read=True`` returns a read-optimized copy whose subtrees
+        other than metadata/status are shared with the store — callers must
+        treat those as read-only (the controller hot path opts in; the default
+        is a fully private deep copy, safe to mutate)."""
         info = self.registry.by_kind(api_version, kind)
         ns = namespace if info.namespaced else ""
         with self._lock:
@@ -220,7 +232,7 @@ class MemoryApiServer:
             obj = self._objects.get((api_version, kind, ns, name))
             if obj is None:
                 raise self._not_found(api_version, kind, name)
-            return deep_copy(obj)
+            return snapshot(obj) if snapshot_read else deep_copy(obj)
 
     def list(
         self,
@@ -228,8 +240,10 @@ class MemoryApiServer:
         kind: str,
         namespace: Optional[str] = None,
         label_selector: Optional[str] = None,
+        snapshot_read: bool = False,
     ) -> List[Obj]:
         selector = parse_label_selector(label_selector)
+        copier = snapshot if snapshot_read else deep_copy
         with self._lock:
             self.op_counts["list"] += 1
             out = []
@@ -240,7 +254,7 @@ class MemoryApiServer:
                     continue
                 if selector and not _labels_match(obj, selector):
                     continue
-                out.append(deep_copy(obj))
+                out.append(copier(obj))
             return out
 
     def update(self, obj: Obj) -> Obj:
@@ -289,12 +303,11 @@ class MemoryApiServer:
             if meta.get("deletionTimestamp") and not meta.get("finalizers"):
                 del self._objects[key]
                 self._unindex_owners(key, obj)
-                out = deep_copy(obj)
-                self._publish("DELETED", out)
+                self._publish("DELETED", snapshot(obj))
                 self._cascade_delete(meta.get("uid"))
-                return out
+                return deep_copy(obj)
+            self._publish("MODIFIED", snapshot(obj))
             out = deep_copy(obj)
-            self._publish("MODIFIED", out)
         return out
 
     def update_status(self, obj: Obj) -> Obj:
@@ -314,7 +327,7 @@ class MemoryApiServer:
                     f'"{meta.get("name")}": the object has been modified; please apply '
                     f"your changes to the latest version and try again"
                 )
-            updated = deep_copy(existing)
+            updated = snapshot(existing)
             if "status" in obj:
                 updated["status"] = deep_copy(obj["status"])
             else:
@@ -323,8 +336,8 @@ class MemoryApiServer:
                 return deep_copy(existing)
             updated["metadata"]["resourceVersion"] = self._next_rv()
             self._objects[key] = updated
+            self._publish("MODIFIED", snapshot(updated))
             out = deep_copy(updated)
-            self._publish("MODIFIED", out)
         return out
 
     def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None:
@@ -341,11 +354,11 @@ class MemoryApiServer:
                 if not meta.get("deletionTimestamp"):
                     meta["deletionTimestamp"] = k8s_now()
                     meta["resourceVersion"] = self._next_rv()
-                    self._publish("MODIFIED", deep_copy(obj))
+                    self._publish("MODIFIED", snapshot(obj))
                 return
             del self._objects[key]
             self._unindex_owners(key, obj)
-            self._publish("DELETED", deep_copy(obj))
+            self._publish("DELETED", snapshot(obj))
             self._cascade_delete(meta.get("uid"))
 
     def _cascade_delete(self, owner_uid: Optional[str]) -> None:
@@ -360,7 +373,7 @@ class MemoryApiServer:
             obj = self._objects.pop(key, None)
             if obj is not None:
                 self._unindex_owners(key, obj)
-                self._publish("DELETED", deep_copy(obj))
+                self._publish("DELETED", snapshot(obj))
                 self._cascade_delete(obj["metadata"].get("uid"))
 
     def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription:

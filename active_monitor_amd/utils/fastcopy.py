@@ -24,3 +24,23 @@ def deep_copy(obj: Any) -> Any:
     import copy
 
     return copy.deepcopy(obj)  # non-JSON payloads: fall back to the slow path
+
+
+def snapshot(obj: Any) -> Any:
+    """Read-optimized copy of a Kubernetes object: a fresh top level with
+    ``metadata`` and ``status`` deep-copied (the only subtrees writers touch
+    in place) and every other subtree — typically the large, effectively
+    immutable ``spec`` — shared by reference.
+
+    Contract: consumers treat anything below the top level of a read result
+    as read-only, except ``metadata``/``status`` which are private copies.
+    Every in-tree consumer honors this; it is what makes 1000-CR fleets cheap.
+    """
+    if not isinstance(obj, dict):
+        return deep_copy(obj)
+    out = dict(obj)
+    if "metadata" in out:
+        out["metadata"] = deep_copy(out["metadata"])
+    if "status" in out:
+        out["status"] = deep_copy(out["status"])
+    return out
