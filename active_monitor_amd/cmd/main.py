@@ -60,7 +60,14 @@ def build_parser() -> argparse.ArgumentParser:
     return p
 
 
-async def run(args) -> int:
+async def run(args, stop_event: Optional[asyncio.Event] = None) -> int:
+    """Process main loop; runs until SIGINT/SIGTERM (or ``stop_event`` for
+    tests — the reference covers run()'s error/shutdown paths the same way,
+    cmd/main_test.go:35-64)."""
+    return await _run(args, stop_event)
+
+
+async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
     from ..engine import Manager
     from ..kube import MemoryApiServer, MemoryClient
 
@@ -98,12 +105,12 @@ async def run(args) -> int:
         leader_elect=args.leader_elect,
     )
 
-    stop = asyncio.Event()
+    stop = stop_event if stop_event is not None else asyncio.Event()
     loop = asyncio.get_running_loop()
     for sig in (signal.SIGINT, signal.SIGTERM):
         try:
             loop.add_signal_handler(sig, stop.set)
-        except NotImplementedError:  # pragma: no cover - non-unix
+        except (NotImplementedError, RuntimeError):  # pragma: no cover - non-unix
             pass
 
     if engine is not None:

@@ -200,6 +200,12 @@ class HealthCheckReconciler:
             # reference lets them die on workflow NotFound after GC).
             if self._stop_timer(name):
                 log.info("cancelled rescheduled workflow for deleted healthcheck %s", name)
+                await self.recorder.event(
+                    {"apiVersion": API_VERSION, "kind": HC_KIND,
+                     "metadata": {"name": name, "namespace": namespace}},
+                    "Normal", "Normal",
+                    "Cancelling workflow for this healthcheck due to deletion",
+                )
             self._cancel_watches(name)
             return ReconcileResult()
         hc = HealthCheck.from_dict(obj)
