@@ -2,7 +2,7 @@
 # targets, adapted to the Python stack).
 PY ?= python3
 
-.PHONY: test test-gpu bench run manifests lint fmt docker-build sweep
+.PHONY: test test-gpu bench run manifests native docker-build sweep
 
 test:
 	$(PY) -m pytest tests/ -q -m "not gpu"
@@ -18,6 +18,9 @@ sweep:
 
 run:
 	$(PY) -m active_monitor_amd.cmd.main --backend memory --max-workers 10
+
+native:
+	$(PY) setup.py build_ext --inplace
 
 manifests:
 	$(PY) -m active_monitor_amd.api.crd > config/crd/bases/activemonitor.keikoproj.io_healthchecks.yaml

@@ -55,6 +55,10 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--workflow-engine", choices=["local", "none"], default="local",
                    help="memory backend: execute workflows locally or leave to an "
                         "external engine")
+    p.add_argument("--shard-index", type=int, default=0,
+                   help="this controller's shard (keyspace split by CR-name hash)")
+    p.add_argument("--shard-count", type=int, default=1,
+                   help="total cooperating controller shards")
     p.add_argument("--zap-log-level", default="info",
                    help="log level (debug/info/warn/error)")
     return p
@@ -103,6 +107,8 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         metrics_addr=parse_bind_address(args.metrics_bind_address),
         health_addr=parse_bind_address(args.health_probe_bind_address),
         leader_elect=args.leader_elect,
+        shard_index=args.shard_index,
+        shard_count=args.shard_count,
     )
 
     stop = stop_event if stop_event is not None else asyncio.Event()

@@ -25,7 +25,25 @@ async def _handle(reader: asyncio.StreamReader, writer: asyncio.StreamWriter, ma
             if line in (b"\r\n", b"\n", b""):
                 break
 
-        if path.startswith("/healthz"):
+        if path.startswith("/statusz"):
+            import json
+
+            stats = {}
+            if manager is not None:
+                rec = getattr(manager, "reconciler", None)
+                stats = {
+                    "ready": manager.ready,
+                    "workers": getattr(manager, "max_workers", None),
+                    "shard": [getattr(manager, "shard_index", 0),
+                              getattr(manager, "shard_count", 1)],
+                    "queue_depth": len(getattr(manager, "queue", [])),
+                    "reconciles": getattr(rec, "reconcile_count", 0),
+                    "completed_runs": getattr(rec, "completed_runs", 0),
+                    "active_watches": rec.active_watches() if rec else 0,
+                    "armed_timers": len(getattr(rec, "repeat_timers_by_name", {})),
+                }
+            body, ctype, code = json.dumps(stats).encode(), "application/json", 200
+        elif path.startswith("/healthz"):
             body, ctype, code = b"ok", "text/plain", 200
         elif path.startswith("/readyz"):
             if manager is None or manager.ready:

@@ -44,6 +44,8 @@ class Manager:
         leader_elect: bool = False,
         leader_identity: str = "",
         record_latencies: bool = True,
+        shard_index: int = 0,
+        shard_count: int = 1,
     ):
         self.client = client
         self.max_workers = max_workers
@@ -64,6 +66,13 @@ class Manager:
         self._servers: List[object] = []
         self.record_latencies = record_latencies
         self.latencies: List[float] = []
+        # horizontal scale-out: shard the CR keyspace by stable name hash so
+        # N cooperating controller processes each own a disjoint subset (one
+        # event loop saturates one core — see docs/DESIGN.md scaling model)
+        if not (0 <= shard_index < shard_count):
+            raise ValueError("shard_index must be in [0, shard_count)")
+        self.shard_index = shard_index
+        self.shard_count = shard_count
 
     # -- lifecycle ----------------------------------------------------------
 
@@ -128,15 +137,27 @@ class Manager:
 
     # -- informer -----------------------------------------------------------
 
+    def _owns(self, name: str) -> bool:
+        if self.shard_count <= 1:
+            return True
+        # stable, process-independent hash (builtin hash() is salted)
+        import zlib
+
+        return zlib.crc32(name.encode()) % self.shard_count == self.shard_index
+
     async def _informer(self) -> None:
         self._sub = self.client.watch(API_VERSION, HC_KIND, self.namespace)
         # initial list AFTER subscribing so no event can slip between the two
         for obj in await self.client.list(API_VERSION, HC_KIND, self.namespace):
             meta = obj.get("metadata") or {}
-            await self.queue.add((meta.get("namespace", ""), meta.get("name", "")))
+            name = meta.get("name", "")
+            if self._owns(name):
+                await self.queue.add((meta.get("namespace", ""), name))
         async for ev in self._sub:
             meta = ev["object"].get("metadata") or {}
-            await self.queue.add((meta.get("namespace", ""), meta.get("name", "")))
+            name = meta.get("name", "")
+            if self._owns(name):
+                await self.queue.add((meta.get("namespace", ""), name))
 
     # -- workers ------------------------------------------------------------
 

@@ -366,11 +366,29 @@ class HealthCheckReconciler:
     # Watch: health-check workflow (reference :607-757)
     # ------------------------------------------------------------------
 
+    _POLL_RETRIES = 3
+
     async def _poll_workflow(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
-        wf = await self.client.get(WF_API_VERSION, WF_KIND, namespace, name,
-                                   snapshot_read=True)
-        status = wf.get("status")
-        return status if isinstance(status, dict) else None
+        """One status poll. NotFound propagates (ends the watch, :618-622);
+        transient apiserver errors are retried briefly and then treated as a
+        missed poll — the reference aborts the whole watch on any error,
+        stalling the CR until an external reconcile."""
+        last: Optional[BaseException] = None
+        for attempt in range(self._POLL_RETRIES):
+            try:
+                wf = await self.client.get(WF_API_VERSION, WF_KIND, namespace, name,
+                                           snapshot_read=True)
+                status = wf.get("status")
+                return status if isinstance(status, dict) else None
+            except NotFoundError:
+                raise
+            except asyncio.CancelledError:
+                raise
+            except Exception as e:
+                last = e
+                await asyncio.sleep(0.05 * (attempt + 1))
+        log.warning("workflow poll %s/%s failed transiently: %s", namespace, name, last)
+        return None  # missed poll; the IEB deadline still bounds the watch
 
     async def _wait_next_poll(
         self, ieb: InverseExponentialBackoff, namespace: str, name: str, since: Optional[int]

@@ -145,3 +145,44 @@ def test_mini_fleet_soak(run):
                                msg="second wave on every CR")
 
     run(go(), timeout=90)
+
+
+def test_sharded_managers_partition_the_fleet(run):
+    """N shard managers over one apiserver each own a disjoint CR subset and
+    together cover the whole fleet exactly once."""
+
+    async def go():
+        server = MemoryApiServer()
+        client = MemoryClient(server)
+        engine = ScriptedWorkflowEngine(client, policy=always_succeed)
+        await engine.start()
+        managers = [
+            Manager(client, max_workers=2, shard_index=i, shard_count=3)
+            for i in range(3)
+        ]
+        for m in managers:
+            await m.start()
+        for i in range(60):
+            await client.create(make_hc(name=f"sh-{i:03d}", repeat=3600, timeout=2))
+
+        async def all_ran():
+            objs = await client.list(API_VERSION, "HealthCheck", "health")
+            return all(
+                (o.get("status") or {}).get("successCount", 0) >= 1 for o in objs
+            )
+
+        deadline = asyncio.get_running_loop().time() + 25
+        while asyncio.get_running_loop().time() < deadline:
+            if await all_ran():
+                break
+            await asyncio.sleep(0.05)
+        assert await all_ran(), "sharded fleet did not fully reconcile"
+        # disjoint ownership: each CR reconciled by exactly one shard
+        per_shard = [m.reconciler.completed_runs for m in managers]
+        assert sum(per_shard) == 60
+        assert all(c > 0 for c in per_shard)  # crc32 spreads 60 names over 3
+        for m in managers:
+            await m.stop()
+        await engine.stop()
+
+    run(go(), timeout=45)
