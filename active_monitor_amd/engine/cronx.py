@@ -227,7 +227,9 @@ def seconds_until_next(spec: str, now: Optional[datetime] = None) -> int:
     ``int(next - now) + 1`` seconds, the +1 compensating integer truncation
     (healthcheck_controller.go:251-263)."""
     if now is None:
-        now = datetime.now(timezone.utc)
+        # local time, like robfig's default Schedule.Next(time.Now()) — an
+        # operator writing "0 9 * * *" means 9am on the controller's clock
+        now = datetime.now().astimezone()
     sched = parse_standard(spec)
     delta = (sched.next(now) - now).total_seconds()
     return int(delta) + 1

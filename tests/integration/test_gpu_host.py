@@ -81,3 +81,14 @@ def test_local_workflow_engine_runs_subprocesses(run):
         return wf["status"]["phase"]
 
     assert run(go()) == "Succeeded"
+
+
+def test_native_extension_loaded_on_gpu_host():
+    """On the benchmark host the native hot path must be the one running —
+    no silent pure-Python fallback."""
+    from active_monitor_amd.utils import fastcopy
+
+    assert fastcopy.NATIVE, "_amcore.so not loaded on the GPU host"
+    from active_monitor_amd import _amcore
+
+    assert _amcore.deep_copy({"a": [1]}) == {"a": [1]}

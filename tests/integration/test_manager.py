@@ -186,3 +186,26 @@ def test_sharded_managers_partition_the_fleet(run):
         await engine.stop()
 
     run(go(), timeout=45)
+
+
+def test_idle_fleet_is_quiescent(run):
+    """Armed timers with far-future repeats must not burn reconciles — the
+    controller is event-driven at idle."""
+
+    async def go():
+        async with Env(workers=4) as env:
+            for i in range(40):
+                await env.create_hc(make_hc(name=f"idle-{i}", repeat=3600, timeout=2))
+            rec = env.manager.reconciler
+
+            async def settled():
+                return rec.completed_runs >= 40
+
+            await env.wait_for(settled, timeout=20, msg="initial runs")
+            await asyncio.sleep(1.0)  # let status-update reconciles drain
+            count = rec.reconcile_count
+            await asyncio.sleep(2.0)
+            assert rec.reconcile_count == count, "reconciles while idle"
+            assert len(env.manager.queue) == 0
+
+    run(go(), timeout=45)
