@@ -48,10 +48,19 @@ def format_k8s_time(dt: datetime) -> str:
     return dt.astimezone(timezone.utc).strftime("%Y-%m-%dT%H:%M:%SZ")
 
 
+_PARSE_TIME_CACHE: Dict[str, datetime] = {}
+
+
 def parse_k8s_time(s: Optional[str]) -> Optional[datetime]:
     if not s:
         return None
-    return datetime.strptime(s, "%Y-%m-%dT%H:%M:%SZ").replace(tzinfo=timezone.utc)
+    cached = _PARSE_TIME_CACHE.get(s)
+    if cached is None:
+        cached = datetime.strptime(s, "%Y-%m-%dT%H:%M:%SZ").replace(tzinfo=timezone.utc)
+        if len(_PARSE_TIME_CACHE) > 4096:
+            _PARSE_TIME_CACHE.clear()
+        _PARSE_TIME_CACHE[s] = cached
+    return cached
 
 
 # ---------------------------------------------------------------------------

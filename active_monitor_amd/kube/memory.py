@@ -35,10 +35,20 @@ Obj = Dict[str, Any]
 Key = Tuple[str, str, str, str]  # (apiVersion, kind, namespace, name)
 
 _SUFFIX_ALPHABET = "bcdfghjklmnpqrstvwxz2456789"  # k8s-style name suffix chars
+_suffix_counter = [random.randrange(27 ** 4)]
 
 
 def _rand_suffix(n: int = 5) -> str:
-    return "".join(random.choices(_SUFFIX_ALPHABET, k=n))
+    """Unique k8s-style suffix: a randomly-seeded counter encoded in the
+    apiserver's suffix alphabet — collision-free and far cheaper than
+    per-character randomness on the create hot path."""
+    _suffix_counter[0] += 1
+    v = _suffix_counter[0]
+    out = []
+    for _ in range(n):
+        v, r = divmod(v, len(_SUFFIX_ALPHABET))
+        out.append(_SUFFIX_ALPHABET[r])
+    return "".join(out)
 
 
 def parse_label_selector(selector: Optional[str]) -> Dict[str, str]:
