@@ -55,6 +55,9 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument("--workflow-engine", choices=["local", "none"], default="local",
                    help="memory backend: execute workflows locally or leave to an "
                         "external engine")
+    p.add_argument("--serve-api", default="",
+                   help="memory backend: also serve the store over the "
+                        "Kubernetes REST API at this address (kubectl-able)")
     p.add_argument("--shard-index", type=int, default=0,
                    help="this controller's shard (keyspace split by CR-name hash)")
     p.add_argument("--shard-count", type=int, default=1,
@@ -99,6 +102,13 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
             from ..workflow import LocalWorkflowEngine
 
             engine = LocalWorkflowEngine(client, args.namespace)
+        if args.serve_api:
+            from ..kube.server import ApiServerFrontend
+
+            addr = parse_bind_address(args.serve_api)
+            frontend = ApiServerFrontend(client.server, addr[0], addr[1])
+            await frontend.start()
+            log.info("serving Kubernetes REST API at %s", frontend.url)
 
     manager = Manager(
         client,
@@ -130,6 +140,8 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         await engine.stop()
     if args.backend == "http":
         await client.close()
+    if args.backend == "memory" and args.serve_api:
+        await frontend.stop()
     return 0
 
 

@@ -47,6 +47,12 @@ class ApiServerFrontend:
 
     async def start(self) -> None:
         app = web.Application()
+        # discovery endpoints (enough for kubectl --server=<url>)
+        app.router.add_get("/api", self._discovery_api)
+        app.router.add_get("/apis", self._discovery_apis)
+        app.router.add_get("/version", self._discovery_version)
+        app.router.add_get("/api/{version}", self._discovery_core_resources)
+        app.router.add_get("/apis/{group}/{version}", self._discovery_group_resources)
         app.router.add_route("*", "/api/{version}/{tail:.*}", self._handle_core)
         app.router.add_route("*", "/apis/{group}/{version}/{tail:.*}", self._handle_group)
         self._runner = web.AppRunner(app, access_log=None, shutdown_timeout=1.0)
@@ -61,6 +67,69 @@ class ApiServerFrontend:
             sub.close()  # end open watch streams so cleanup() is immediate
         if self._runner is not None:
             await self._runner.cleanup()
+
+    # -- discovery ----------------------------------------------------------
+
+    def _groups(self):
+        groups = {}
+        for (av, _), info in self.server.registry._by_kind.items():
+            if "/" in av:
+                g, v = av.split("/", 1)
+                groups.setdefault(g, set()).add(v)
+        return groups
+
+    async def _discovery_api(self, request: web.Request) -> web.Response:
+        return web.json_response({"kind": "APIVersions", "versions": ["v1"]})
+
+    async def _discovery_apis(self, request: web.Request) -> web.Response:
+        groups = []
+        for g, versions in sorted(self._groups().items()):
+            gv = sorted({f"{g}/{v}" for v in versions})
+            groups.append({
+                "name": g,
+                "versions": [{"groupVersion": x, "version": x.split("/", 1)[1]} for x in gv],
+                "preferredVersion": {"groupVersion": gv[0], "version": gv[0].split("/", 1)[1]},
+            })
+        return web.json_response({"kind": "APIGroupList", "apiVersion": "v1", "groups": groups})
+
+    async def _discovery_version(self, request: web.Request) -> web.Response:
+        from .. import __version__
+
+        return web.json_response({
+            "major": "1", "minor": "33",
+            "gitVersion": f"v1.33.0-active-monitor-amd+{__version__}",
+        })
+
+    def _resource_list(self, api_version: str) -> dict:
+        resources = []
+        for (av, kind), info in sorted(self.server.registry._by_kind.items()):
+            if av != api_version:
+                continue
+            resources.append({
+                "name": info.plural,
+                "singularName": kind.lower(),
+                "namespaced": info.namespaced,
+                "kind": kind,
+                "verbs": ["create", "delete", "get", "list", "patch", "update", "watch"],
+            })
+            if (av, kind) == ("activemonitor.keikoproj.io/v1alpha1", "HealthCheck"):
+                resources[-1]["shortNames"] = ["hc", "hcs"]
+                resources.append({
+                    "name": f"{info.plural}/status",
+                    "singularName": "",
+                    "namespaced": True,
+                    "kind": kind,
+                    "verbs": ["get", "update", "patch"],
+                })
+        return {"kind": "APIResourceList", "apiVersion": "v1",
+                "groupVersion": api_version, "resources": resources}
+
+    async def _discovery_core_resources(self, request: web.Request) -> web.Response:
+        return web.json_response(self._resource_list(request.match_info["version"]))
+
+    async def _discovery_group_resources(self, request: web.Request) -> web.Response:
+        gv = f'{request.match_info["group"]}/{request.match_info["version"]}'
+        return web.json_response(self._resource_list(gv))
 
     # -- routing -----------------------------------------------------------
 

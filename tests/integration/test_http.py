@@ -146,3 +146,39 @@ def test_full_controller_over_http(run):
                 await engine.stop()
 
     run(go(), timeout=40)
+
+
+def test_discovery_endpoints_for_kubectl(run):
+    """/api, /apis, /version and per-group APIResourceLists — enough for
+    `kubectl --server=<frontend-url>` to resolve `hc` and list resources."""
+    import aiohttp
+
+    async def go():
+        async with HttpEnv() as env:
+            async with aiohttp.ClientSession() as s:
+                async def get(path):
+                    async with s.get(env.frontend.url + path) as r:
+                        assert r.status == 200, path
+                        return await r.json()
+
+                api = await get("/api")
+                assert api["versions"] == ["v1"]
+
+                apis = await get("/apis")
+                names = {g["name"] for g in apis["groups"]}
+                assert {"activemonitor.keikoproj.io", "argoproj.io",
+                        "rbac.authorization.k8s.io"} <= names
+
+                core = await get("/api/v1")
+                core_names = {r["name"] for r in core["resources"]}
+                assert {"serviceaccounts", "events", "namespaces"} <= core_names
+
+                am = await get("/apis/activemonitor.keikoproj.io/v1alpha1")
+                hc = next(r for r in am["resources"] if r["name"] == "healthchecks")
+                assert hc["shortNames"] == ["hc", "hcs"]
+                assert any(r["name"] == "healthchecks/status" for r in am["resources"])
+
+                ver = await get("/version")
+                assert "gitVersion" in ver
+
+    run(go(), timeout=30)
