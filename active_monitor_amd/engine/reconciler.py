@@ -324,6 +324,7 @@ class HealthCheckReconciler:
         except Exception as e:
             await self._event(hc, "Warning", "Error creating or submitting workflow")
             raise e
+        await self._event(hc, "Normal", "workflow is parsed from healthcheck")
         wf = {
             "apiVersion": WF_API_VERSION,
             "kind": WF_KIND,
@@ -347,6 +348,7 @@ class HealthCheckReconciler:
         except Exception as e:
             await self._event(hc, "Warning", "Error creating or submitting remedyworkflow")
             raise e
+        await self._event(hc, "Normal", "Remedy workflow is parsed from healthcheck")
         wf = {
             "apiVersion": WF_API_VERSION,
             "kind": WF_KIND,
