@@ -212,7 +212,16 @@ def main():
     if world > 1:
         import torch.distributed as dist
 
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("AM_DIST_BACKEND", "")
+        if not backend:
+            # RCCL when every rank has its own GPU; otherwise gloo (the
+            # workload is CPU control-plane — the collective is only the
+            # barrier + result gather)
+            backend = (
+                "nccl"
+                if torch.cuda.is_available() and torch.cuda.device_count() >= world
+                else "gloo"
+            )
         if backend == "nccl":
             torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", "0")))
         dist.init_process_group(backend=backend)
