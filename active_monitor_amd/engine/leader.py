@@ -9,8 +9,6 @@ from __future__ import annotations
 import asyncio
 import logging
 import time
-from typing import Optional
-
 from ..api.types import k8s_now, parse_k8s_time
 from ..kube.client import KubeClient
 from ..kube.errors import AlreadyExistsError, ConflictError, NotFoundError

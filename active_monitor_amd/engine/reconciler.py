@@ -29,8 +29,8 @@ from __future__ import annotations
 import asyncio
 import logging
 import time
-from dataclasses import dataclass, field
-from typing import Any, Dict, Optional, Set, Tuple
+from dataclasses import dataclass
+from typing import Any, Dict, Optional, Set
 
 from .. import API_VERSION
 from ..api.types import HealthCheck, k8s_now, parse_k8s_time

@@ -33,7 +33,7 @@ import threading
 from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 from prometheus_client import CollectorRegistry, Gauge, generate_latest
-from prometheus_client.core import CounterMetricFamily, Metric
+from prometheus_client.core import Metric
 
 log = logging.getLogger("active_monitor_amd.metrics")
 
