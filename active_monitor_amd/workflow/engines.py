@@ -49,12 +49,22 @@ def never_complete(wf: Dict[str, Any]) -> PolicyResult:
 
 
 class _EngineBase:
-    def __init__(self, client: KubeClient, namespace: Optional[str] = None):
+    """``ttl_seconds`` deletes completed workflows after the given delay —
+    the role Argo's ``ttlStrategy.secondsAfterCompletion`` plays for the
+    reference (its install configures 1800s, deploy-argo.yaml:1162-1173);
+    None disables (tests that inspect completed workflows)."""
+
+    DEFAULT_TTL = 1800.0
+
+    def __init__(self, client: KubeClient, namespace: Optional[str] = None,
+                 ttl_seconds: Optional[float] = DEFAULT_TTL):
         self.client = client
         self.namespace = namespace
+        self.ttl_seconds = ttl_seconds
         self._task: Optional[asyncio.Task] = None
         self._sub = None
         self._inflight: Dict[str, asyncio.Task] = {}
+        self._ttl_handles: list = []
 
     async def start(self) -> None:
         self._sub = self.client.watch(WF_API_VERSION, WF_KIND, self.namespace)
@@ -71,6 +81,9 @@ class _EngineBase:
                 pass
         for t in self._inflight.values():
             t.cancel()
+        for h in self._ttl_handles:
+            h.cancel()
+        self._ttl_handles.clear()
 
     async def _loop(self) -> None:
         async for ev in self._sub:
@@ -86,6 +99,31 @@ class _EngineBase:
 
     async def _run(self, wf: Dict[str, Any]) -> None:  # pragma: no cover - abstract
         raise NotImplementedError
+
+    def _schedule_ttl(self, wf: Dict[str, Any]) -> None:
+        if self.ttl_seconds is None:
+            return
+        meta = wf["metadata"]
+        ns, name = meta.get("namespace", ""), meta["name"]
+
+        def _gc() -> None:
+            async def delete() -> None:
+                try:
+                    await self.client.delete(WF_API_VERSION, WF_KIND, ns, name)
+                except Exception:
+                    pass
+
+            task = asyncio.ensure_future(delete())
+            self._inflight[f"ttl:{ns}/{name}"] = task
+            task.add_done_callback(
+                lambda t: self._inflight.pop(f"ttl:{ns}/{name}", None)
+            )
+
+        loop = asyncio.get_event_loop()
+        self._ttl_handles.append(loop.call_later(self.ttl_seconds, _gc))
+        if len(self._ttl_handles) > 512:  # drop spent handles
+            self._ttl_handles = [h for h in self._ttl_handles if not h.cancelled()
+                                 and h.when() > loop.time()]
 
     async def _set_status(self, wf: Dict[str, Any], status: Dict[str, Any]) -> None:
         """Write the Workflow status the way the Argo controller does (the
@@ -113,8 +151,9 @@ class ScriptedWorkflowEngine(_EngineBase):
         policy: Policy = always_succeed,
         delay: float = 0.0,
         namespace: Optional[str] = None,
+        ttl_seconds: Optional[float] = _EngineBase.DEFAULT_TTL,
     ):
-        super().__init__(client, namespace)
+        super().__init__(client, namespace, ttl_seconds)
         self.policy = policy
         self.delay = delay
         self.completed = 0
@@ -133,6 +172,7 @@ class ScriptedWorkflowEngine(_EngineBase):
         if len(decision) > 2 and decision[2]:  # type: ignore[misc]
             status["outputs"] = decision[2]  # type: ignore[misc]
         await self._set_status(wf, status)
+        self._schedule_ttl(wf)
         self.completed += 1
 
 
@@ -155,8 +195,9 @@ class LocalWorkflowEngine(_EngineBase):
     README.md:275-285).
     """
 
-    def __init__(self, client: KubeClient, namespace: Optional[str] = None):
-        super().__init__(client, namespace)
+    def __init__(self, client: KubeClient, namespace: Optional[str] = None,
+                 ttl_seconds: Optional[float] = _EngineBase.DEFAULT_TTL):
+        super().__init__(client, namespace, ttl_seconds)
         self.completed = 0
 
     async def _run(self, wf: Dict[str, Any]) -> None:
@@ -182,6 +223,7 @@ class LocalWorkflowEngine(_EngineBase):
         if outputs:
             status["outputs"] = {"parameters": outputs}
         await self._set_status(wf, status)
+        self._schedule_ttl(wf)
         self.completed += 1
 
     async def _run_spec(self, spec: Dict[str, Any]) -> Tuple[Phase, str, List[Dict[str, Any]]]:

@@ -1,0 +1,127 @@
+#!/usr/bin/env python3
+"""Steady-state soak: a real timer-driven fleet (no bench waves).
+
+Unlike bench.py — which drives explicit waves for deterministic step timing —
+this runs the controller exactly as production would: CRs with short
+``repeatAfterSec``, repeats fired by the controller's own timers, for a fixed
+wall-clock duration. Reports sustained cycles/s, completion-latency
+percentiles, and RSS growth (leak canary).
+
+Usage: python benchmarks/soak.py [--crs 1000] [--repeat 5] [--duration 120]
+"""
+import argparse
+import asyncio
+import json
+import os
+import resource
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def rss_mb() -> float:
+    """Current (not peak) resident set size."""
+    try:
+        with open("/proc/self/status") as f:
+            for line in f:
+                if line.startswith("VmRSS:"):
+                    return int(line.split()[1]) / 1024.0
+    except OSError:
+        pass
+    return resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024.0
+
+
+async def main_async(args) -> dict:
+    from active_monitor_amd.engine import Manager
+    from active_monitor_amd.kube import MemoryApiServer, MemoryClient
+    from active_monitor_amd.workflow import ScriptedWorkflowEngine
+
+    server = MemoryApiServer()
+    client = MemoryClient(server)
+    engine = ScriptedWorkflowEngine(client, policy=lambda wf: ("Succeeded", ""),
+                                    ttl_seconds=args.ttl)
+    await engine.start()
+    manager = Manager(client, max_workers=args.workers)
+    await manager.start()
+
+    inline = (
+        "apiVersion: argoproj.io/v1alpha1\nkind: Workflow\nspec:\n"
+        "  entrypoint: e\n  templates:\n    - name: e\n      container:\n"
+        "        command: [echo, ok]\n"
+    )
+    for i in range(args.crs):
+        await client.create({
+            "apiVersion": "activemonitor.keikoproj.io/v1alpha1",
+            "kind": "HealthCheck",
+            "metadata": {"name": f"soak-{i:05d}", "namespace": "health"},
+            "spec": {
+                "repeatAfterSec": args.repeat,
+                "level": "cluster",
+                "workflow": {
+                    "generateName": f"soak-{i:05d}-wf-",
+                    "workflowtimeout": max(args.repeat, 5),
+                    "resource": {
+                        "namespace": "health",
+                        "serviceAccount": f"soak-sa-{i % 8}",
+                        "source": {"inline": inline},
+                    },
+                },
+            },
+        })
+
+    rec = manager.reconciler
+    # settle: every CR has run once
+    while rec.completed_runs < args.crs:
+        await asyncio.sleep(0.05)
+    rss_start = rss_mb()
+    runs_start = rec.completed_runs
+    t0 = time.monotonic()
+    samples = []
+    while time.monotonic() - t0 < args.duration:
+        await asyncio.sleep(min(5.0, args.duration / 10))
+        samples.append({
+            "t": round(time.monotonic() - t0, 1),
+            "completed": rec.completed_runs - runs_start,
+            "rss_mb": round(rss_mb(), 1),
+            "active_watches": rec.active_watches(),
+            "queue": len(manager.queue),
+            "objects": len(server),
+        })
+    elapsed = time.monotonic() - t0
+    total = rec.completed_runs - runs_start
+    lat = manager.drain_latencies()
+    result = {
+        "crs": args.crs,
+        "repeat_after_sec": args.repeat,
+        "duration_s": round(elapsed, 1),
+        "cycles": total,
+        "cycles_per_sec": round(total / elapsed, 1),
+        "expected_cycles_per_sec": round(args.crs / args.repeat, 1),
+        "p50_reconcile_ms": round(statistics.median(lat) * 1000, 4) if lat else None,
+        "rss_start_mb": round(rss_start, 1),
+        "rss_end_mb": round(rss_mb(), 1),
+        "store_objects_end": len(server),
+        "samples": samples,
+    }
+    await manager.stop()
+    await engine.stop()
+    return result
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--crs", type=int, default=1000)
+    ap.add_argument("--repeat", type=int, default=5)
+    ap.add_argument("--workers", type=int, default=8)
+    ap.add_argument("--duration", type=float, default=120.0)
+    ap.add_argument("--ttl", type=float, default=30.0,
+                    help="completed-workflow TTL (Argo ttlStrategy equivalent)")
+    args = ap.parse_args()
+    print(json.dumps(asyncio.run(main_async(args))))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
