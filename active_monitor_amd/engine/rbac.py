@@ -91,9 +91,38 @@ class RBACProvisioner:
     (the reference threads a typed clientset through each helper so unit
     tests can pass a fake — here the client itself is injectable)."""
 
-    def __init__(self, client: KubeClient, recorder: Optional[EventRecorder] = None):
+    #: how long a successful ensure suppresses re-checking the same object.
+    #: The reference re-gets every RBAC object on every cycle (:333-408); with
+    #: a fleet at 1 Hz that is ~4 redundant apiserver round-trips per cycle.
+    #: External deletions are still repaired within this window; 0 disables.
+    ENSURE_TTL = 30.0
+
+    def __init__(self, client: KubeClient, recorder: Optional[EventRecorder] = None,
+                 ensure_ttl: Optional[float] = None):
         self.client = client
         self.recorder = recorder
+        self.ensure_ttl = self.ENSURE_TTL if ensure_ttl is None else ensure_ttl
+        self._ensured: Dict[tuple, float] = {}
+
+    def _fresh(self, key: tuple) -> bool:
+        import time
+
+        if self.ensure_ttl <= 0:
+            return False
+        exp = self._ensured.get(key)
+        return exp is not None and exp > time.monotonic()
+
+    def _mark(self, key: tuple) -> None:
+        import time
+
+        if self.ensure_ttl > 0:
+            if len(self._ensured) > 50000:
+                now = time.monotonic()
+                self._ensured = {k: v for k, v in self._ensured.items() if v > now}
+            self._ensured[key] = time.monotonic() + self.ensure_ttl
+
+    def _invalidate(self, kind: str, namespace: str, name: str) -> None:
+        self._ensured.pop((kind, namespace, name), None)
 
     async def _event(self, hc_obj: Dict[str, Any], ev_type: str, message: str) -> None:
         if self.recorder is not None:
@@ -102,8 +131,12 @@ class RBACProvisioner:
     # -- creates (get-then-create, reuse if present) -----------------------
 
     async def create_service_account(self, name: str, namespace: str) -> str:
+        cache_key = ("ServiceAccount", namespace, name)
+        if self._fresh(cache_key):
+            return name
         try:
             sa = await self.client.get("v1", "ServiceAccount", namespace, name)
+            self._mark(cache_key)
             return sa["metadata"]["name"]
         except NotFoundError:
             pass
@@ -117,12 +150,18 @@ class RBACProvisioner:
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
+            self._mark(cache_key)
             return name
+        self._mark(cache_key)
         return created["metadata"]["name"]
 
     async def create_cluster_role(self, name: str, rules: List[PolicyRule]) -> str:
+        cache_key = ("ClusterRole", "", name)
+        if self._fresh(cache_key):
+            return name
         try:
             cr = await self.client.get(RBAC_API_VERSION, "ClusterRole", "", name)
+            self._mark(cache_key)
             return cr["metadata"]["name"]
         except NotFoundError:
             pass
@@ -137,14 +176,20 @@ class RBACProvisioner:
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
+            self._mark(cache_key)
             return name
+        self._mark(cache_key)
         return created["metadata"]["name"]
 
     async def create_cluster_role_binding(
         self, name: str, role_name: str, sa_name: str, sa_namespace: str
     ) -> str:
+        cache_key = ("ClusterRoleBinding", "", name)
+        if self._fresh(cache_key):
+            return name
         try:
             crb = await self.client.get(RBAC_API_VERSION, "ClusterRoleBinding", "", name)
+            self._mark(cache_key)
             return crb["metadata"]["name"]
         except NotFoundError:
             pass
@@ -166,14 +211,20 @@ class RBACProvisioner:
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
+            self._mark(cache_key)
             return name
+        self._mark(cache_key)
         return created["metadata"]["name"]
 
     async def create_namespace_role(
         self, name: str, namespace: str, rules: List[PolicyRule]
     ) -> str:
+        cache_key = ("Role", namespace, name)
+        if self._fresh(cache_key):
+            return name
         try:
             role = await self.client.get(RBAC_API_VERSION, "Role", namespace, name)
+            self._mark(cache_key)
             return role["metadata"]["name"]
         except NotFoundError:
             pass
@@ -188,14 +239,20 @@ class RBACProvisioner:
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
+            self._mark(cache_key)
             return name
+        self._mark(cache_key)
         return created["metadata"]["name"]
 
     async def create_namespace_role_binding(
         self, name: str, role_name: str, sa_name: str, namespace: str
     ) -> str:
+        cache_key = ("RoleBinding", namespace, name)
+        if self._fresh(cache_key):
+            return name
         try:
             rb = await self.client.get(RBAC_API_VERSION, "RoleBinding", namespace, name)
+            self._mark(cache_key)
             return rb["metadata"]["name"]
         except NotFoundError:
             pass
@@ -217,7 +274,9 @@ class RBACProvisioner:
         except AlreadyExistsError:
             # concurrent reconciles sharing an SA race get-then-create;
             # losing the race means the object exists — reuse it
+            self._mark(cache_key)
             return name
+        self._mark(cache_key)
         return created["metadata"]["name"]
 
     # -- deletes (only objects carrying the managed-by label) --------------
@@ -238,18 +297,23 @@ class RBACProvisioner:
                 return
 
     async def delete_service_account(self, name: str, namespace: str) -> None:
+        self._invalidate("ServiceAccount", namespace, name)
         await self._delete_if_managed("v1", "ServiceAccount", namespace, name)
 
     async def delete_cluster_role(self, name: str) -> None:
+        self._invalidate("ClusterRole", "", name)
         await self._delete_if_managed(RBAC_API_VERSION, "ClusterRole", "", name)
 
     async def delete_cluster_role_binding(self, name: str) -> None:
+        self._invalidate("ClusterRoleBinding", "", name)
         await self._delete_if_managed(RBAC_API_VERSION, "ClusterRoleBinding", "", name)
 
     async def delete_namespace_role(self, name: str, namespace: str) -> None:
+        self._invalidate("Role", namespace, name)
         await self._delete_if_managed(RBAC_API_VERSION, "Role", namespace, name)
 
     async def delete_namespace_role_binding(self, name: str, namespace: str) -> None:
+        self._invalidate("RoleBinding", namespace, name)
         await self._delete_if_managed(RBAC_API_VERSION, "RoleBinding", namespace, name)
 
     # -- orchestration ------------------------------------------------------

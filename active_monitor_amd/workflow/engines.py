@@ -64,7 +64,9 @@ class _EngineBase:
         self._task: Optional[asyncio.Task] = None
         self._sub = None
         self._inflight: Dict[str, asyncio.Task] = {}
-        self._ttl_handles: list = []
+        from collections import deque
+
+        self._ttl_handles: "deque" = deque()
 
     async def start(self) -> None:
         self._sub = self.client.watch(WF_API_VERSION, WF_KIND, self.namespace)
@@ -121,9 +123,13 @@ class _EngineBase:
 
         loop = asyncio.get_event_loop()
         self._ttl_handles.append(loop.call_later(self.ttl_seconds, _gc))
-        if len(self._ttl_handles) > 512:  # drop spent handles
-            self._ttl_handles = [h for h in self._ttl_handles if not h.cancelled()
-                                 and h.when() > loop.time()]
+        # handles are appended in firing order (constant ttl): drop spent
+        # ones from the front — O(1) amortized
+        now = loop.time()
+        while self._ttl_handles and (
+            self._ttl_handles[0].cancelled() or self._ttl_handles[0].when() <= now
+        ):
+            self._ttl_handles.popleft()
 
     async def _set_status(self, wf: Dict[str, Any], status: Dict[str, Any]) -> None:
         """Write the Workflow status the way the Argo controller does (the
