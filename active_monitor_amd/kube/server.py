@@ -151,16 +151,11 @@ class ApiServerFrontend:
         #   namespaces/{ns}/{plural}/{name}/status
         parts = [p for p in tail.split("/") if p]
         namespace = ""
-        if len(parts) >= 2 and parts[0] == "namespaces" and (
-            len(parts) > 2 or request.method in ("GET", "DELETE")
-        ):
-            # disambiguate /api/v1/namespaces/{name} (the Namespace resource)
-            # from /api/v1/namespaces/{ns}/{plural}...
-            if len(parts) == 2:
-                parts = ["namespaces", parts[1]]  # the Namespace object itself
-            else:
-                namespace = parts[1]
-                parts = parts[2:]
+        # /api/v1/namespaces/{ns}/{plural}... is namespaced access;
+        # /api/v1/namespaces[/{name}] (≤2 segments) is the Namespace resource
+        if len(parts) >= 3 and parts[0] == "namespaces":
+            namespace = parts[1]
+            parts = parts[2:]
         plural = parts[0] if parts else ""
         name = parts[1] if len(parts) > 1 else ""
         subresource = parts[2] if len(parts) > 2 else ""
