@@ -171,7 +171,8 @@ class HttpClient:
 
     async def _request(self, method: str, path: str, body: Optional[Obj] = None,
                        params: Optional[Dict[str, str]] = None) -> Obj:
-        assert self._session is not None, "call start() first"
+        if self._session is None:
+            raise RuntimeError("HttpClient.start() must be called before requests")
         async with self._session.request(
             method, self.base_url + path,
             json=body if body is not None else None, params=params,

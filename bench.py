@@ -61,7 +61,6 @@ spec:
 def make_cr(i: int, ns: str, cron_frac: float, remedy_frac: float):
     """Synthetic CR mix per BASELINE config 5: mixed cron + repeatAfterSec
     with Remedy."""
-    n_total = 1.0
     name = f"hc-{i:05d}"
     is_remedy = (i % 100) < remedy_frac * 100
     is_cron = not is_remedy and (i % 100) < (remedy_frac + cron_frac) * 100

@@ -69,8 +69,6 @@ def test_metric_values_after_cycles(run):
 
 
 def test_failure_metrics_and_error_message(run):
-    from active_monitor_amd.workflow import always_fail
-
     async def go():
         async with Env(policy=lambda wf: ("Failed", "probe exploded")) as env:
             await env.create_hc(make_hc(name="failing", repeat=1, timeout=2))
