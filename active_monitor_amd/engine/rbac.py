@@ -99,9 +99,14 @@ class RBACProvisioner:
 
     def __init__(self, client: KubeClient, recorder: Optional[EventRecorder] = None,
                  ensure_ttl: Optional[float] = None):
+        import os
+
         self.client = client
         self.recorder = recorder
-        self.ensure_ttl = self.ENSURE_TTL if ensure_ttl is None else ensure_ttl
+        if ensure_ttl is None:
+            env = os.environ.get("AM_RBAC_ENSURE_TTL")
+            ensure_ttl = float(env) if env else self.ENSURE_TTL
+        self.ensure_ttl = ensure_ttl
         self._ensured: Dict[tuple, float] = {}
 
     def _fresh(self, key: tuple) -> bool:
