@@ -49,7 +49,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="restrict watch to one namespace (default: all)")
     p.add_argument("--backend", choices=["memory", "http"], default="memory")
     p.add_argument("--server", default="",
-                   help="apiserver URL for --backend http")
+                   help="apiserver URL for --backend http (default: in-cluster "
+                        "config, then $KUBECONFIG/~/.kube/config)")
+    p.add_argument("--kubeconfig", default=None,
+                   help="explicit kubeconfig path for --backend http")
     p.add_argument("--token", default="", help="bearer token for --backend http")
     p.add_argument("--insecure-skip-tls-verify", action="store_true")
     p.add_argument("--workflow-engine", choices=["local", "none"], default="local",
@@ -89,12 +92,13 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
 
     engine = None
     if args.backend == "http":
-        from ..kube.http import HttpClient
+        from ..kube.config import get_config
 
-        client = HttpClient(
-            args.server, token=args.token or None,
-            verify=not args.insecure_skip_tls_verify,
+        cfg = get_config(
+            server=args.server, token=args.token,
+            insecure=args.insecure_skip_tls_verify, kubeconfig=args.kubeconfig,
         )
+        client = cfg.make_client()
         await client.start()
     else:
         client = MemoryClient(MemoryApiServer())

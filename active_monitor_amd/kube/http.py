@@ -133,6 +133,8 @@ class HttpClient:
         self.verify = verify
         self.ca_cert = ca_cert
         self.registry = registry
+        #: optional (cert_path, key_path) for client-certificate auth
+        self.client_cert: tuple = (None, None)
         self._session: Optional[aiohttp.ClientSession] = None
 
     async def start(self) -> None:
@@ -147,6 +149,11 @@ class HttpClient:
                 ssl_ctx.verify_mode = ssl.CERT_NONE
             elif self.ca_cert:
                 ssl_ctx = ssl.create_default_context(cafile=self.ca_cert)
+            cert, key = self.client_cert
+            if cert:
+                if ssl_ctx is None:
+                    ssl_ctx = ssl.create_default_context()
+                ssl_ctx.load_cert_chain(cert, key)
         self._session = aiohttp.ClientSession(
             headers=headers, connector=aiohttp.TCPConnector(ssl=ssl_ctx)
         )
