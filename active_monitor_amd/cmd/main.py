@@ -100,6 +100,12 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         )
         client = cfg.make_client()
         await client.start()
+        try:
+            await client.ping()
+        except Exception as e:
+            log.error("cannot reach apiserver %s: %s", cfg.server, e)
+            await client.close()
+            return 1
     else:
         client = MemoryClient(MemoryApiServer())
         if args.workflow_engine == "local":

@@ -162,6 +162,12 @@ class HttpClient:
         if self._session is not None:
             await self._session.close()
 
+    async def ping(self) -> None:
+        """Fail fast when the apiserver is unreachable (used at startup, the
+        run()-returns-error-for-invalid-config contract of the reference's
+        cmd/main_test.go:35-49)."""
+        await self._request("GET", "/version")
+
     # -- paths --------------------------------------------------------------
 
     def _collection_path(self, api_version: str, kind: str, namespace: Optional[str]) -> str:

@@ -76,3 +76,20 @@ def test_run_disabled_endpoints(run):
         assert await asyncio.wait_for(task, 15) == 0
 
     run(go(), timeout=30)
+
+
+def test_run_http_backend_unreachable_server_errors(run):
+    """Invalid apiserver config → run() returns an error rather than running
+    blind (reference cmd/main_test.go:35-49)."""
+
+    async def go():
+        args = build_parser().parse_args([
+            "--backend", "http",
+            "--server", "http://127.0.0.1:1",  # nothing listens there
+            "--metrics-bind-address", "0",
+            "--health-probe-bind-address", "0",
+        ])
+        rc = await asyncio.wait_for(cli_run(args, asyncio.Event()), 30)
+        assert rc == 1
+
+    run(go(), timeout=40)
