@@ -1,5 +1,6 @@
 """Cluster config resolution tests (GetConfigOrDie equivalent)."""
 import base64
+from pathlib import Path
 
 import pytest
 import yaml
@@ -35,7 +36,7 @@ def test_token_auth(tmp_path):
     assert cfg.server == "https://api.example:6443"
     assert cfg.token == "sekret"
     assert cfg.verify is True
-    assert open(cfg.ca_cert_path, "rb").read() == b"CA PEM"
+    assert Path(cfg.ca_cert_path).read_bytes() == b"CA PEM"
 
 
 def test_client_cert_auth(tmp_path):
@@ -44,8 +45,8 @@ def test_client_cert_auth(tmp_path):
         "client-key-data": base64.b64encode(b"KEY").decode(),
     })
     cfg = load_kubeconfig(str(p))
-    assert open(cfg.client_cert_path, "rb").read() == b"CERT"
-    assert open(cfg.client_key_path, "rb").read() == b"KEY"
+    assert Path(cfg.client_cert_path).read_bytes() == b"CERT"
+    assert Path(cfg.client_key_path).read_bytes() == b"KEY"
     client = cfg.make_client()
     assert client.client_cert[0] == cfg.client_cert_path
 
