@@ -65,7 +65,10 @@ class Manager:
         self._stopped = False
         self._servers: List[object] = []
         self.record_latencies = record_latencies
-        self.latencies: List[float] = []
+        from collections import deque
+
+        # bounded: an unbounded list leaks ~8B/reconcile on week-long fleets
+        self.latencies: "deque" = deque(maxlen=200_000)
         # horizontal scale-out: shard the CR keyspace by stable name hash so
         # N cooperating controller processes each own a disjoint subset (one
         # event loop saturates one core — see docs/DESIGN.md scaling model)
@@ -192,6 +195,6 @@ class Manager:
     # -- bench/test helpers --------------------------------------------------
 
     def drain_latencies(self) -> List[float]:
-        out = self.latencies
-        self.latencies = []
+        out = list(self.latencies)
+        self.latencies.clear()
         return out
