@@ -111,11 +111,21 @@ class HealthCheckReconciler:
     # timers
     # ------------------------------------------------------------------
 
-    def get_timer_by_name(self, name: str) -> Optional[RepeatTimer]:
-        return self.repeat_timers_by_name.get(name)
+    def get_timer_by_name(self, name: str, namespace: str = "") -> Optional[RepeatTimer]:
+        """Timers are keyed (namespace, name) — the reference keys by name
+        alone (:139), making same-name CRs in different namespaces fight over
+        one timer slot; fixed here with no other visible change. The
+        name-only form falls back to any-namespace lookup for API parity."""
+        if namespace:
+            return self.repeat_timers_by_name.get((namespace, name))
+        for (ns, n), entry in self.repeat_timers_by_name.items():
+            if n == name:
+                return entry
+        return None
 
     def _arm_repeat_timer(self, name: str, namespace: str, delay: float) -> None:
-        old = self.repeat_timers_by_name.get(name)
+        key = (namespace, name)
+        old = self.repeat_timers_by_name.get(key)
         if old is not None:
             old.stop()
         entry = RepeatTimer()
@@ -123,7 +133,7 @@ class HealthCheckReconciler:
         entry.handle = loop.call_later(
             max(0.0, delay), self._on_timer_fired, entry, name, namespace
         )
-        self.repeat_timers_by_name[name] = entry
+        self.repeat_timers_by_name[key] = entry
 
     def _on_timer_fired(self, entry: RepeatTimer, name: str, namespace: str) -> None:
         entry.fired = True
@@ -132,8 +142,8 @@ class HealthCheckReconciler:
         # Synchronous enqueue — no fire-and-forget task to lose.
         self.queue.add_nowait((namespace, name), {TIMER_FLAG})
 
-    def _stop_timer(self, name: str) -> bool:
-        entry = self.repeat_timers_by_name.pop(name, None)
+    def _stop_timer(self, name: str, namespace: str = "") -> bool:
+        entry = self.repeat_timers_by_name.pop((namespace, name), None)
         if entry is not None:
             entry.stop()
             return True
@@ -143,24 +153,24 @@ class HealthCheckReconciler:
     # watch-task registry
     # ------------------------------------------------------------------
 
-    def _spawn_watch(self, name: str, coro) -> asyncio.Task:
+    def _spawn_watch(self, key, coro) -> asyncio.Task:
         task = asyncio.get_running_loop().create_task(coro)
-        self._watch_tasks.setdefault(name, set()).add(task)
+        self._watch_tasks.setdefault(key, set()).add(task)
 
-        def _done(t: asyncio.Task, name=name) -> None:
-            tasks = self._watch_tasks.get(name)
+        def _done(t: asyncio.Task, key=key) -> None:
+            tasks = self._watch_tasks.get(key)
             if tasks is not None:
                 tasks.discard(t)
                 if not tasks:
-                    self._watch_tasks.pop(name, None)
+                    self._watch_tasks.pop(key, None)
             if not t.cancelled() and t.exception() is not None:
-                log.error("watch task for %s failed: %s", name, t.exception())
+                log.error("watch task for %s failed: %s", key, t.exception())
 
         task.add_done_callback(_done)
         return task
 
-    def _cancel_watches(self, name: str) -> None:
-        for t in list(self._watch_tasks.get(name, ())):
+    def _cancel_watches(self, key) -> None:
+        for t in list(self._watch_tasks.get(key, ())):
             t.cancel()
 
     def active_watches(self) -> int:
@@ -198,7 +208,7 @@ class HealthCheckReconciler:
             # CR deleted: stop the repeat timer so self-scheduling halts
             # (:180-184); in-flight watches are cancelled proactively (the
             # reference lets them die on workflow NotFound after GC).
-            if self._stop_timer(name):
+            if self._stop_timer(name, namespace):
                 log.info("cancelled rescheduled workflow for deleted healthcheck %s", name)
                 await self.recorder.event(
                     {"apiVersion": API_VERSION, "kind": HC_KIND,
@@ -206,7 +216,7 @@ class HealthCheckReconciler:
                     "Normal", "Normal",
                     "Cancelling workflow for this healthcheck due to deletion",
                 )
-            self._cancel_watches(name)
+            self._cancel_watches((namespace, name))
             return ReconcileResult()
         hc = HealthCheck.from_dict(obj)
         return await self._process_or_recover(hc, from_timer=TIMER_FLAG in flags)
@@ -260,7 +270,7 @@ class HealthCheckReconciler:
             )
             await self.update_healthcheck_status(hc)
             return ReconcileResult()
-        elif not from_timer and self._watch_tasks.get(hc.name):
+        elif not from_timer and self._watch_tasks.get((hc.namespace, hc.name)):
             # a run for this CR is already in flight: spurious reconciles
             # (informer list/watch overlap at startup, spec edits mid-run)
             # must not submit a duplicate workflow. The reference double-
@@ -281,13 +291,13 @@ class HealthCheckReconciler:
             if (
                 not from_timer
                 and int(time.time() - finished_unix) < hc.spec.repeat_after_sec
-                and self.get_timer_by_name(hc.name) is not None
+                and self.get_timer_by_name(hc.name, hc.namespace) is not None
             ):
                 return ReconcileResult()
         elif (
             not from_timer
             and int(time.time() - finished_unix) < spec.repeat_after_sec
-            and self.get_timer_by_name(hc.name) is not None
+            and self.get_timer_by_name(hc.name, hc.namespace) is not None
         ):
             # already executed recently and a repeat is scheduled (:264-267)
             return ReconcileResult()
@@ -301,7 +311,8 @@ class HealthCheckReconciler:
         generated_name = await self.create_submit_workflow(hc)
         # non-blocking watch: the reconcile worker is freed immediately
         self._spawn_watch(
-            hc.name, self.watch_workflow_reschedule(wf_namespace, generated_name, hc)
+            (hc.namespace, hc.name),
+            self.watch_workflow_reschedule(wf_namespace, generated_name, hc),
         )
         return ReconcileResult()
 
@@ -573,11 +584,11 @@ class HealthCheckReconciler:
         try:
             await self.update_healthcheck_status(hc)
         except NotFoundError:
-            self._stop_timer(hc.name)
+            self._stop_timer(hc.name, hc.namespace)
             return
         except Exception as e:
             await self._event(hc, "Warning", "Error updating healthcheck resource")
-            self._stop_timer(hc.name)
+            self._stop_timer(hc.name, hc.namespace)
             raise e
         await self._event(hc, "Normal", "Rescheduled workflow for next run")
 
@@ -706,7 +717,7 @@ class HealthCheckReconciler:
     # ------------------------------------------------------------------
 
     def stop_all(self) -> None:
-        for name in list(self.repeat_timers_by_name):
-            self._stop_timer(name)
-        for name in list(self._watch_tasks):
-            self._cancel_watches(name)
+        for (ns, name) in list(self.repeat_timers_by_name):
+            self._stop_timer(name, ns)
+        for key in list(self._watch_tasks):
+            self._cancel_watches(key)

@@ -335,3 +335,39 @@ def test_custom_metrics_wired_on_success(run):
             ) == 42
 
     run(go(), timeout=30)
+
+
+def test_same_name_crs_in_different_namespaces(run):
+    """Same-name CRs in two namespaces run independently — the reference's
+    name-only timer map makes them fight over one slot (its :139); fixed by
+    keying timers and watches by (namespace, name)."""
+
+    async def go():
+        async with Env(policy=always_succeed) as env:
+            for ns in ("health", "default"):
+                await env.create_hc(make_hc(name="dup", ns=ns, repeat=1, timeout=2))
+
+            async def both_repeat():
+                a = await env.client.get(
+                    "activemonitor.keikoproj.io/v1alpha1", "HealthCheck",
+                    "health", "dup")
+                b = await env.client.get(
+                    "activemonitor.keikoproj.io/v1alpha1", "HealthCheck",
+                    "default", "dup")
+                return (
+                    (a.get("status") or {}).get("successCount", 0) >= 2
+                    and (b.get("status") or {}).get("successCount", 0) >= 2
+                )
+
+            deadline = asyncio.get_running_loop().time() + 25
+            while asyncio.get_running_loop().time() < deadline:
+                if await both_repeat():
+                    break
+                await asyncio.sleep(0.1)
+            assert await both_repeat(), "one namespace starved the other"
+            rec = env.manager.reconciler
+            assert rec.get_timer_by_name("dup", "health") is not None
+            assert rec.get_timer_by_name("dup", "default") is not None
+            assert rec.get_timer_by_name("dup", "health") is not rec.get_timer_by_name("dup", "default")
+
+    run(go(), timeout=45)
