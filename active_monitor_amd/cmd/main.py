@@ -67,6 +67,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="total cooperating controller shards")
     p.add_argument("--zap-log-level", default="info",
                    help="log level (debug/info/warn/error)")
+    p.add_argument("--gc-threshold", type=int, default=50000,
+                   help="gen-0 GC threshold; large fleets hold 100k+ live "
+                        "objects and default CPython thresholds cost ~30%% "
+                        "throughput (0 keeps the interpreter default)")
     return p
 
 
@@ -78,6 +82,10 @@ async def run(args, stop_event: Optional[asyncio.Event] = None) -> int:
 
 
 async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
+    if getattr(args, "gc_threshold", 0):
+        import gc
+
+        gc.set_threshold(args.gc_threshold, 50, 50)
     from ..engine import Manager
     from ..kube import MemoryApiServer, MemoryClient
 

@@ -111,6 +111,9 @@ async def main_async(args) -> dict:
 
 
 def main() -> int:
+    import gc
+
+    gc.set_threshold(50000, 50, 50)
     ap = argparse.ArgumentParser()
     ap.add_argument("--crs", type=int, default=1000)
     ap.add_argument("--repeat", type=int, default=5)
