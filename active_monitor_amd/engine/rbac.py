@@ -337,11 +337,13 @@ class RBACProvisioner:
         wf_remedy_namespace = ""
         if not hc.spec.remedy_workflow.is_empty():
             if hc.spec.remedy_workflow.resource is None:
-                await self._event(hc_obj, "Warning", "RemedyWorkflow is set but Resource is nil")
+                await self._event(hc.to_dict(), "Warning",
+                                  "RemedyWorkflow is set but Resource is nil")
                 raise ValueError("RemedyWorkflow is set but Resource is nil")
             if hc.spec.remedy_workflow.resource.service_account == "":
                 await self._event(
-                    hc_obj, "Warning", "ServiceAccount for the RemedyWorkflow is not specified"
+                    hc.to_dict(), "Warning",
+                    "ServiceAccount for the RemedyWorkflow is not specified"
                 )
                 raise ValueError("ServiceAccount for the RemedyWorkflow is not specified")
             if hc_sa == hc.spec.remedy_workflow.resource.service_account:
@@ -389,7 +391,7 @@ class RBACProvisioner:
                     wf_remedy_namespace,
                 )
         else:
-            await self._event(hc_obj, "Warning", "level is not set")
+            await self._event(hc.to_dict(), "Warning", "level is not set")
             raise ValueError("level is not set")
 
     async def delete_rbac_for_workflow(self, hc: HealthCheck) -> None:

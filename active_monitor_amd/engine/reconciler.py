@@ -190,7 +190,17 @@ class HealthCheckReconciler:
     # ------------------------------------------------------------------
 
     async def _event(self, hc: HealthCheck, ev_type: str, message: str) -> None:
-        await self.recorder.event(hc.to_dict(), ev_type, ev_type, message)
+        # the recorder only reads object identity — don't serialize the spec
+        involved = {
+            "apiVersion": API_VERSION,
+            "kind": HC_KIND,
+            "metadata": {
+                "name": hc.metadata.name,
+                "namespace": hc.metadata.namespace,
+                "uid": hc.metadata.uid,
+            },
+        }
+        await self.recorder.event(involved, ev_type, ev_type, message)
 
     # ------------------------------------------------------------------
     # Reconcile entry (reference :170-223)
