@@ -168,8 +168,10 @@ class ScriptedWorkflowEngine(_EngineBase):
         decision = self.policy(wf)
         if decision is None:
             return  # leave pending: the controller's IEB timeout takes over
-        await self._set_status(wf, {"phase": "Running", "startedAt": _now_iso()})
         if self.delay > 0:
+            # instant completions skip the intermediate Running write — one
+            # status update (and one spurious watcher wakeup) less per run
+            await self._set_status(wf, {"phase": "Running", "startedAt": _now_iso()})
             await asyncio.sleep(self.delay)
         phase, message = decision[0], decision[1]
         status: Dict[str, Any] = {"phase": phase, "finishedAt": _now_iso()}
