@@ -93,3 +93,82 @@ def test_run_http_backend_unreachable_server_errors(run):
         assert rc == 1
 
     run(go(), timeout=40)
+
+
+def test_standalone_stack_serve_api_with_amctl(run, capsys, tmp_path):
+    """The full standalone product: CLI controller (memory backend + local
+    subprocess engine + served REST API) driven by amctl; a health check
+    executes end-to-end."""
+    import yaml
+
+    from active_monitor_amd.cmd.ctl import build_parser as ctl_parser
+    from active_monitor_amd.cmd.ctl import run as ctl_run
+
+    async def go():
+        args = build_parser().parse_args([
+            "--backend", "memory",
+            "--serve-api", "127.0.0.1:18281",
+            "--metrics-bind-address", "0",
+            "--health-probe-bind-address", "0",
+            "--max-workers", "4",
+        ])
+        stop = asyncio.Event()
+        task = asyncio.ensure_future(cli_run(args, stop))
+        url = "http://127.0.0.1:18281"
+
+        async def ctl(*argv):
+            rc = await ctl_run(ctl_parser().parse_args(["--server", url, *argv]))
+            return rc, capsys.readouterr().out
+
+        # wait for the API to come up
+        import aiohttp
+        deadline = asyncio.get_running_loop().time() + 15
+        async with aiohttp.ClientSession() as s:
+            while asyncio.get_running_loop().time() < deadline:
+                try:
+                    async with s.get(url + "/version"):
+                        break
+                except aiohttp.ClientError:
+                    await asyncio.sleep(0.1)
+
+        doc = {
+            "apiVersion": "activemonitor.keikoproj.io/v1alpha1",
+            "kind": "HealthCheck",
+            "metadata": {"name": "standalone", "namespace": "health"},
+            "spec": {
+                "repeatAfterSec": 60, "level": "cluster",
+                "workflow": {
+                    "generateName": "standalone-wf-",
+                    "workflowtimeout": 20,
+                    "resource": {
+                        "namespace": "health", "serviceAccount": "sa",
+                        "source": {"inline": (
+                            "spec:\n  entrypoint: main\n  templates:\n"
+                            "    - name: main\n      container:\n"
+                            "        command: [\"true\"]\n"
+                        )},
+                    },
+                },
+            },
+        }
+        f = tmp_path / "hc.yaml"
+        f.write_text(yaml.safe_dump(doc))
+        rc, out = await ctl("apply", "-f", str(f))
+        assert rc == 0 and "created" in out
+
+        # the local engine executes `true` and the controller records success
+        deadline = asyncio.get_running_loop().time() + 25
+        ok = False
+        while asyncio.get_running_loop().time() < deadline:
+            rc, out = await ctl("get", "hc", "standalone", "-n", "health")
+            line = next(l for l in out.splitlines() if l.startswith("standalone"))
+            if line.split()[1] == "Succeeded":
+                ok = True
+                break
+            await asyncio.sleep(0.2)
+        assert ok, f"standalone check never succeeded: {out}"
+
+        stop.set()
+        assert await asyncio.wait_for(task, 15) == 0
+
+    run(go(), timeout=60)
