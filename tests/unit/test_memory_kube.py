@@ -190,3 +190,19 @@ def test_error_helpers():
     assert ignore_not_found(e) is e
     assert is_storage_error(RuntimeError("StorageError: invalid object in etcd"))
     assert not is_storage_error(RuntimeError("other"))
+
+
+def test_update_without_resource_version_is_last_write_wins():
+    """Updates omitting resourceVersion skip the conflict check (apiserver
+    semantics: rv-less update = last write wins)."""
+    s = MemoryApiServer()
+    s.create(hc())
+    a = s.get("activemonitor.keikoproj.io/v1alpha1", "HealthCheck", "health", "check-1")
+    b = s.get("activemonitor.keikoproj.io/v1alpha1", "HealthCheck", "health", "check-1")
+    a["spec"]["repeatAfterSec"] = 10
+    s.update(a)
+    b["spec"]["repeatAfterSec"] = 20
+    del b["metadata"]["resourceVersion"]
+    s.update(b)  # no conflict despite being stale
+    final = s.get("activemonitor.keikoproj.io/v1alpha1", "HealthCheck", "health", "check-1")
+    assert final["spec"]["repeatAfterSec"] == 20

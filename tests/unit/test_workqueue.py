@@ -121,3 +121,22 @@ def test_get_waits_for_add(run):
         return item[0]
 
     assert run(go()) == "k"
+
+
+def test_delayed_add_earlier_item_reschedules_waker(run):
+    """An earlier-firing delayed item added after a later one must still fire
+    first (the waker reschedules)."""
+
+    async def go():
+        q = WorkQueue()
+        q.add_after_nowait("later", 0.5)
+        await asyncio.sleep(0.05)
+        q.add_after_nowait("earlier", 0.1)
+        first = await asyncio.wait_for(q.get(), 2)
+        assert first[0] == "earlier"
+        q.done_nowait("earlier")
+        second = await asyncio.wait_for(q.get(), 2)
+        assert second[0] == "later"
+        q.done_nowait("later")
+
+    run(go())
