@@ -176,7 +176,6 @@ class Manager:
             try:
                 result = await self.reconciler.reconcile(ns, name, flags)
             except asyncio.CancelledError:
-                await self.queue.done(key)
                 raise
             except Exception as e:  # reconciler already guards; belt & braces
                 log.error("worker %d: reconcile %s/%s raised: %s", idx, ns, name, e)
