@@ -592,7 +592,7 @@ class HealthCheckReconciler:
                 pass  # keep the submit-time value
         self._arm_repeat_timer(hc.name, hc.namespace, repeat_after_sec)
         try:
-            await self.update_healthcheck_status(hc)
+            await self.update_healthcheck_status(hc, fresh=fresh)
         except NotFoundError:
             self._stop_timer(hc.name, hc.namespace)
             return
@@ -701,24 +701,31 @@ class HealthCheckReconciler:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):
             return
-        await self.update_healthcheck_status(hc)
+        await self.update_healthcheck_status(hc, fresh=fresh)
 
     # ------------------------------------------------------------------
     # Status persistence (reference :1445-1462)
     # ------------------------------------------------------------------
 
-    async def update_healthcheck_status(self, hc: HealthCheck, retries: int = 5) -> None:
-        """Fresh Get + status-subresource update with conflict retry."""
+    async def update_healthcheck_status(
+        self, hc: HealthCheck, retries: int = 5,
+        fresh: Optional[Dict[str, Any]] = None,
+    ) -> None:
+        """Fresh Get + status-subresource update with conflict retry.
+        ``fresh`` lets callers that just read the object donate that read for
+        the first attempt (one round-trip saved per completed run)."""
         last: Optional[BaseException] = None
         for attempt in range(retries):
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
-                                          snapshot_read=True)
+            if fresh is None:
+                fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace,
+                                              hc.name, snapshot_read=True)
             fresh["status"] = hc.status.to_dict()
             try:
                 await self.client.update_status(fresh)
                 return
             except ConflictError as e:
                 last = e
+                fresh = None  # stale: re-read on the next attempt
                 await asyncio.sleep(0.01 * (attempt + 1))
         raise last if last else RuntimeError("status update failed")
 
