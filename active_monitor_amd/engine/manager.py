@@ -46,6 +46,7 @@ class Manager:
         record_latencies: bool = True,
         shard_index: int = 0,
         shard_count: int = 1,
+        enable_wf_hub: bool = True,
     ):
         self.client = client
         self.max_workers = max_workers
@@ -76,6 +77,10 @@ class Manager:
             raise ValueError("shard_index must be in [0, shard_count)")
         self.shard_index = shard_index
         self.shard_count = shard_count
+        # event-driven completion wakeups; False falls back to the reference's
+        # pure inverse-exponential polling (same semantics, higher latency)
+        self.enable_wf_hub = enable_wf_hub
+        self.wf_hub = None
 
     # -- lifecycle ----------------------------------------------------------
 
@@ -99,11 +104,12 @@ class Manager:
                 self, health=self.health_addr, metrics=self.metrics_addr
             )
 
-        from .watchhub import WorkflowWatchHub
+        if self.enable_wf_hub:
+            from .watchhub import WorkflowWatchHub
 
-        self.wf_hub = WorkflowWatchHub(self.client, self.namespace)
-        await self.wf_hub.start()
-        self.reconciler.wf_hub = self.wf_hub
+            self.wf_hub = WorkflowWatchHub(self.client, self.namespace)
+            await self.wf_hub.start()
+            self.reconciler.wf_hub = self.wf_hub
 
         self._tasks.append(asyncio.ensure_future(self._informer()))
         for i in range(self.max_workers):

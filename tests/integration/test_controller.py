@@ -371,3 +371,35 @@ def test_same_name_crs_in_different_namespaces(run):
             assert rec.get_timer_by_name("dup", "health") is not rec.get_timer_by_name("dup", "default")
 
     run(go(), timeout=45)
+
+
+def test_pure_polling_without_watch_hub(run):
+    """With the watch hub disabled the controller degrades to the reference's
+    pure IEB polling and still completes cycles (higher detection latency,
+    identical semantics)."""
+    from active_monitor_amd.engine import Manager
+    from active_monitor_amd.kube import MemoryApiServer, MemoryClient
+
+    async def go():
+        client = MemoryClient(MemoryApiServer())
+        engine = ScriptedWorkflowEngine(client, policy=always_succeed)
+        await engine.start()
+        manager = Manager(client, max_workers=2, enable_wf_hub=False)
+        await manager.start()
+        assert manager.reconciler.wf_hub is None
+        try:
+            await client.create(make_hc(name="polled", repeat=1, timeout=4))
+            deadline = asyncio.get_running_loop().time() + 25
+            ok = False
+            while asyncio.get_running_loop().time() < deadline:
+                obj = await client.get(API_VERSION, "HealthCheck", "health", "polled")
+                if (obj.get("status") or {}).get("successCount", 0) >= 2:
+                    ok = True
+                    break
+                await asyncio.sleep(0.1)
+            assert ok, "no cycles completed on the pure-polling path"
+        finally:
+            await manager.stop()
+            await engine.stop()
+
+    run(go(), timeout=45)
