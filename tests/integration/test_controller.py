@@ -403,3 +403,22 @@ def test_pure_polling_without_watch_hub(run):
             await engine.stop()
 
     run(go(), timeout=45)
+
+
+def test_repeat_takes_precedence_over_cron(run):
+    """When both repeatAfterSec>0 and schedule.cron are set, the interval wins
+    and the cron is ignored (the reference's branch order, :238-263)."""
+
+    async def go():
+        async with Env(policy=always_succeed) as env:
+            # cron says every hour; repeat says every second — repeat must win
+            await env.create_hc(make_hc(name="both", repeat=1, cron="0 * * * *",
+                                        timeout=2))
+
+            async def ran_twice():
+                hc = await env.get_hc("both")
+                return hc.status.success_count >= 2
+
+            await env.wait_for(ran_twice, timeout=20, msg="interval-driven repeats")
+
+    run(go(), timeout=40)

@@ -126,3 +126,37 @@ def test_cron_next_is_strictly_future_and_matching(expr, now):
     # idempotence: asking again from just before nxt returns nxt
     again = sched.next(nxt - timedelta(seconds=1))
     assert again == nxt
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    st.integers(min_value=0, max_value=10**5),
+    st.integers(min_value=0, max_value=10**5),
+    st.sampled_from(["", "0.1", "0.5", "0.99", "garbage"]),
+    st.integers(min_value=0, max_value=10**6),
+)
+def test_backoff_params_invariants(bmax, bmin, factor, timeout):
+    from active_monitor_amd.engine.backoff import compute_backoff_params
+
+    mx, mn, f, to = compute_backoff_params(bmax, bmin, factor, timeout)
+    # defaults floor at 1s; explicit values pass through verbatim
+    assert mx == (bmax if bmax else max(timeout // 2, 1))
+    assert mn == (bmin if bmin else max(timeout // 60, 1))
+    assert 0 < f  # factor is 0.5 or the parsed value
+    assert to == timeout
+
+
+@settings(max_examples=150, deadline=None)
+@given(st.dictionaries(
+    st.text(alphabet="abc-.x/", min_size=1, max_size=12),
+    st.text(alphabet="abcXYZ0-_.", max_size=12),
+    max_size=4,
+))
+def test_label_selector_roundtrip(labels):
+    from active_monitor_amd.kube.memory import parse_label_selector
+
+    if not labels:
+        assert parse_label_selector("") == {}
+        return
+    sel = ",".join(f"{k}={v}" for k, v in labels.items())
+    assert parse_label_selector(sel) == labels
