@@ -67,6 +67,9 @@ def build_parser() -> argparse.ArgumentParser:
                    help="total cooperating controller shards")
     p.add_argument("--zap-log-level", default="info",
                    help="log level (debug/info/warn/error)")
+    p.add_argument("--log-format", choices=["console", "json"], default="console",
+                   help="console or structured JSON lines (the reference's "
+                        "zap production encoding)")
     p.add_argument("--gc-threshold", type=int, default=50000,
                    help="gen-0 GC threshold; large fleets hold 100k+ live "
                         "objects and default CPython thresholds cost ~30%% "
@@ -92,10 +95,17 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
     level = {"debug": logging.DEBUG, "info": logging.INFO,
              "warn": logging.WARNING, "error": logging.ERROR}.get(
         args.zap_log_level, logging.INFO)
-    logging.basicConfig(
-        level=level,
-        format="%(asctime)s %(levelname)s %(name)s %(message)s",
-    )
+    if getattr(args, "log_format", "console") == "json":
+        from .logfmt import JsonFormatter
+
+        handler = logging.StreamHandler()
+        handler.setFormatter(JsonFormatter())
+        logging.basicConfig(level=level, handlers=[handler], force=True)
+    else:
+        logging.basicConfig(
+            level=level,
+            format="%(asctime)s %(levelname)s %(name)s %(message)s",
+        )
     log = logging.getLogger("active_monitor_amd.main")
 
     engine = None

@@ -113,3 +113,26 @@ def test_cli_parse_bind_address():
     assert args.health_probe_bind_address == ":8081"
     assert args.max_workers == 10
     assert args.leader_elect is False
+
+
+def test_json_log_formatter():
+    import json as _json
+    import logging
+
+    from active_monitor_amd.cmd.logfmt import JsonFormatter
+
+    rec = logging.LogRecord("active_monitor_amd.x", logging.WARNING, "f.py", 1,
+                            "reconcile %s failed", ("hc-1",), None)
+    out = _json.loads(JsonFormatter().format(rec))
+    assert out["level"] == "warning"
+    assert out["logger"] == "active_monitor_amd.x"
+    assert out["msg"] == "reconcile hc-1 failed"
+    assert isinstance(out["ts"], float)
+
+
+def test_cli_log_format_flag():
+    from active_monitor_amd.cmd.main import build_parser
+
+    args = build_parser().parse_args(["--log-format", "json"])
+    assert args.log_format == "json"
+    assert build_parser().parse_args([]).log_format == "console"
