@@ -16,7 +16,7 @@ from typing import Optional
 
 from aiohttp import web
 
-from .errors import ApiError
+from .errors import ApiError, InvalidError
 from .memory import MemoryApiServer
 
 log = logging.getLogger("active_monitor_amd.kube.server")
@@ -182,15 +182,28 @@ class ApiServerFrontend:
             if request.method == "GET":
                 return web.json_response(self.server.get(api_version, kind, namespace, name))
             if request.method == "POST":
-                obj = await request.json()
+                try:
+                    obj = await request.json()
+                except Exception:
+                    return web.json_response(
+                        _status_body(InvalidError("request body is not valid JSON")),
+                        status=400,
+                    )
                 meta = obj.setdefault("metadata", {})
                 if namespace and not meta.get("namespace"):
                     meta["namespace"] = namespace
                 return web.json_response(self.server.create(obj), status=201)
-            if request.method == "PUT" and subresource == "status":
-                return web.json_response(self.server.update_status(await request.json()))
             if request.method == "PUT":
-                return web.json_response(self.server.update(await request.json()))
+                try:
+                    body = await request.json()
+                except Exception:
+                    return web.json_response(
+                        _status_body(InvalidError("request body is not valid JSON")),
+                        status=400,
+                    )
+                if subresource == "status":
+                    return web.json_response(self.server.update_status(body))
+                return web.json_response(self.server.update(body))
             if request.method == "DELETE":
                 self.server.delete(api_version, kind, namespace, name)
                 return web.json_response({"kind": "Status", "status": "Success"})

@@ -182,3 +182,30 @@ def test_discovery_endpoints_for_kubectl(run):
                 assert "gitVersion" in ver
 
     run(go(), timeout=30)
+
+
+def test_http_error_paths(run):
+    """Frontend negative paths: unknown resources 404 with a k8s Status body,
+    malformed JSON bodies surface as errors, unsupported methods 405."""
+    import aiohttp
+
+    async def go():
+        async with HttpEnv() as env:
+            async with aiohttp.ClientSession() as s:
+                base = env.frontend.url
+                async with s.get(base + "/apis/nope.example/v1/widgets") as r:
+                    assert r.status == 404
+                    body = await r.json()
+                    assert body["kind"] == "Status" and body["status"] == "Failure"
+                async with s.post(
+                    base + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/health/healthchecks",
+                    data=b"{not json", headers={"Content-Type": "application/json"},
+                ) as r:
+                    assert r.status == 400
+                    assert (await r.json())["kind"] == "Status"
+                async with s.patch(
+                    base + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/health/healthchecks/x"
+                ) as r:
+                    assert r.status in (404, 405)
+
+    run(go(), timeout=30)
