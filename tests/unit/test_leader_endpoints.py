@@ -136,3 +136,45 @@ def test_cli_log_format_flag():
     args = build_parser().parse_args(["--log-format", "json"])
     assert args.log_format == "json"
     assert build_parser().parse_args([]).log_format == "console"
+
+
+def test_statusz_endpoint(run):
+    class FakeRec:
+        reconcile_count = 7
+        completed_runs = 5
+        repeat_timers_by_name = {("health", "a"): object()}
+
+        def active_watches(self):
+            return 2
+
+    class FakeManager:
+        ready = True
+        max_workers = 4
+        shard_index = 1
+        shard_count = 3
+        queue = []
+        reconciler = FakeRec()
+
+    async def go():
+        servers = await serve_endpoints(FakeManager(), health=("127.0.0.1", 0))
+        try:
+            port = servers[0].sockets[0].getsockname()[1]
+            body, code = await asyncio.get_running_loop().run_in_executor(
+                None, _fetch, f"http://127.0.0.1:{port}/statusz"
+            )
+            assert code == 200
+            import json
+
+            stats = json.loads(body)
+            assert stats["ready"] is True
+            assert stats["workers"] == 4
+            assert stats["shard"] == [1, 3]
+            assert stats["reconciles"] == 7
+            assert stats["completed_runs"] == 5
+            assert stats["active_watches"] == 2
+            assert stats["armed_timers"] == 1
+        finally:
+            for s in servers:
+                s.close()
+
+    run(go())
