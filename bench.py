@@ -206,6 +206,12 @@ def main():
     args = parse_args()
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.gpus != world and rank == 0:
+        print(
+            f"note: --gpus {args.gpus} but WORLD_SIZE={world}; N>1 must be "
+            "launched via torch.distributed.run (reporting n_gpus=WORLD_SIZE)",
+            file=sys.stderr,
+        )
 
     try:
         import torch
