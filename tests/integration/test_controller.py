@@ -422,3 +422,45 @@ def test_repeat_takes_precedence_over_cron(run):
             await env.wait_for(ran_twice, timeout=20, msg="interval-driven repeats")
 
     run(go(), timeout=40)
+
+
+def test_pause_then_resume_lifecycle(run):
+    """Active → paused (repeatAfterSec:0) → Stopped; then resumed → runs
+    again. Covers live spec edits across schedule states."""
+
+    async def go():
+        async with Env(policy=always_succeed) as env:
+            await env.create_hc(make_hc(name="toggle", repeat=1, timeout=2))
+
+            async def ran():
+                hc = await env.get_hc("toggle")
+                return hc.status.success_count >= 1
+
+            await env.wait_for(ran, msg="initial run")
+
+            # pause it
+            obj = await env.client.get(API_VERSION, "HealthCheck", "health", "toggle")
+            obj["spec"].pop("repeatAfterSec", None)
+            await env.client.update(obj)
+
+            async def stopped():
+                hc = await env.get_hc("toggle")
+                return hc.status.status == "Stopped"
+
+            await env.wait_for(stopped, timeout=15, msg="paused → Stopped")
+            count_at_pause = (await env.get_hc("toggle")).status.success_count
+            await asyncio.sleep(2.0)
+            assert (await env.get_hc("toggle")).status.success_count == count_at_pause
+
+            # resume it
+            obj = await env.client.get(API_VERSION, "HealthCheck", "health", "toggle")
+            obj["spec"]["repeatAfterSec"] = 1
+            await env.client.update(obj)
+
+            async def resumed():
+                hc = await env.get_hc("toggle")
+                return hc.status.success_count > count_at_pause
+
+            await env.wait_for(resumed, timeout=15, msg="resumed and running")
+
+    run(go(), timeout=60)
