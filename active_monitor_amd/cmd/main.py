@@ -161,7 +161,21 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         await engine.start()
     await manager.start()
     log.info("starting manager: workers=%d backend=%s", args.max_workers, args.backend)
-    await stop.wait()
+    # run until signalled — or until the manager reports a fatal condition
+    # (lost leadership lease): the reference treats losing the lease as fatal
+    # for the replica (cmd/main.go:87-88), so a deposed process exits nonzero
+    # instead of reconciling without the lease.
+    stop_w = asyncio.ensure_future(stop.wait())
+    fatal_w = asyncio.ensure_future(manager.fatal.wait())
+    done, pending = await asyncio.wait(
+        {stop_w, fatal_w}, return_when=asyncio.FIRST_COMPLETED
+    )
+    for t in pending:
+        t.cancel()
+    rc = 0
+    if fatal_w in done:
+        log.error("manager fatal: %s", manager.fatal_reason)
+        rc = 1
     log.info("shutting down")
     await manager.stop()
     if engine is not None:
@@ -170,7 +184,7 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         await client.close()
     if args.backend == "memory" and args.serve_api:
         await frontend.stop()
-    return 0
+    return rc
 
 
 def main(argv=None) -> int:

@@ -5,7 +5,9 @@ from .manager import Manager
 from .parse import (
     WorkflowParseError,
     parse_remedy_workflow_from_healthcheck,
+    parse_remedy_workflow_from_healthcheck_async,
     parse_workflow_from_healthcheck,
+    parse_workflow_from_healthcheck_async,
 )
 from .rbac import (
     DEFAULT_HEALTHCHECK_RULES,
@@ -32,8 +34,10 @@ __all__ = [
     "compute_backoff_params",
     "parse_go_duration",
     "parse_remedy_workflow_from_healthcheck",
+    "parse_remedy_workflow_from_healthcheck_async",
     "parse_standard",
     "parse_workflow_from_healthcheck",
+    "parse_workflow_from_healthcheck_async",
     "resolve_rbac_rules",
     "seconds_until_next",
 ]

@@ -65,6 +65,11 @@ class Manager:
         self._started = asyncio.Event()
         self._stopped = False
         self._servers: List[object] = []
+        # set when the manager must die (e.g. lost leadership lease); the
+        # entrypoint awaits it alongside the stop signal — in the reference
+        # losing the lease is fatal for the replica (cmd/main.go:87-88)
+        self.fatal = asyncio.Event()
+        self.fatal_reason: Optional[str] = None
         self.record_latencies = record_latencies
         from collections import deque
 
@@ -95,7 +100,10 @@ class Manager:
                 identity=self.leader_identity or f"manager-{id(self):x}",
             )
             await elector.acquire()
-            self._tasks.append(asyncio.ensure_future(elector.renew_loop()))
+            renew = asyncio.ensure_future(elector.renew_loop())
+            renew.add_done_callback(self._renew_done)
+            self._tasks.append(renew)
+            self.elector = elector
 
         if self.health_addr is not None or self.metrics_addr is not None:
             from .endpoints import serve_endpoints
@@ -140,9 +148,20 @@ class Manager:
         for srv in self._servers:
             srv.close()
 
+    def _renew_done(self, task: "asyncio.Task") -> None:
+        """A finished renew loop means the lease is gone: a deposed replica
+        must stop reconciling immediately or two actives double-submit
+        workflows (split-brain). Surface it as a fatal manager condition."""
+        if task.cancelled() or self._stopped:
+            return
+        exc = task.exception()
+        self.fatal_reason = str(exc) if exc else "leader-election renew loop exited"
+        log.error("leadership lost: %s — shutting down", self.fatal_reason)
+        self.fatal.set()
+
     @property
     def ready(self) -> bool:
-        return self._started.is_set() and not self._stopped
+        return self._started.is_set() and not self._stopped and not self.fatal.is_set()
 
     # -- informer -----------------------------------------------------------
 

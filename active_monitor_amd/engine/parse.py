@@ -22,12 +22,13 @@ parseRemedyWorkflowFromHealthcheck (healthcheck_controller.go:876-1125):
 """
 from __future__ import annotations
 
-from typing import Any, Dict, Tuple
+import asyncio
+from typing import Any, Dict, Optional, Tuple
 
 import yaml
 
 from ..api.types import HealthCheck
-from ..store import get_artifact_reader
+from ..store import ArtifactReadError, get_artifact_reader
 
 WF_INSTANCE_ID_LABEL_KEY = "workflows.argoproj.io/controller-instanceid"
 WF_INSTANCE_ID = "activemonitor-workflows"
@@ -95,14 +96,48 @@ def _validated_spec(data: Dict[str, Any], remedy: bool) -> Dict[str, Any]:
     return spec_raw
 
 
-def parse_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, Any], Dict[str, str]]:
+#: ceiling on one offloaded artifact read before the reconcile errors out
+ARTIFACT_READ_TIMEOUT = 35.0
+
+
+def _source_blocks(resource) -> bool:
+    """URL and file sources do real I/O; inline is a memory read."""
+    src = getattr(resource, "source", None)
+    return src is not None and (src.url is not None or src.file is not None)
+
+
+async def _read_source_async(resource) -> bytes:
+    """Read an artifact source without blocking the event loop.
+
+    The reference's URL fetch blocks only its own goroutine (url.go:20-57);
+    a single-loop design must offload it. Inline sources stay on-loop (pure
+    memory); URL/file reads run on a worker thread under a hard timeout so one
+    hung source cannot stall other CRs' reconciles, watches or timers."""
+    if resource is None:
+        return b""
+    reader = get_artifact_reader(resource.source)
+    if not _source_blocks(resource):
+        return reader.read()
+    try:
+        return await asyncio.wait_for(asyncio.to_thread(reader.read), ARTIFACT_READ_TIMEOUT)
+    except asyncio.TimeoutError:
+        raise ArtifactReadError(
+            f"artifact read timed out after {ARTIFACT_READ_TIMEOUT:.0f}s"
+        ) from None
+
+
+def parse_workflow_from_healthcheck(
+    hc: HealthCheck, content: Optional[bytes] = None
+) -> Tuple[Dict[str, Any], Dict[str, str]]:
     """Returns ``(spec_dict, labels)`` for the health-check workflow, mutating
     ``hc.spec.workflow.timeout`` when it defaults from RepeatAfterSec
-    (reference :980-995)."""
-    content = b""
-    if hc.spec.workflow.resource is not None:
-        reader = get_artifact_reader(hc.spec.workflow.resource.source)
-        content = reader.read()
+    (reference :980-995). ``content`` short-circuits the artifact read (used
+    by the async variant, which offloads blocking sources to a thread)."""
+    if content is None:
+        content = b""
+        if hc.spec.workflow.resource is not None:
+            reader = get_artifact_reader(hc.spec.workflow.resource.source)
+            content = reader.read()
     data = _decode(content, remedy=False)
     labels = _extract_labels(data)
     spec = _validated_spec(data, remedy=False)
@@ -122,14 +157,26 @@ def parse_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, Any], Di
     return spec, labels
 
 
-def parse_remedy_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, Any], Dict[str, str]]:
+async def parse_workflow_from_healthcheck_async(
+    hc: HealthCheck,
+) -> Tuple[Dict[str, Any], Dict[str, str]]:
+    """Event-loop-safe variant: blocking sources (URL/file) are read on a
+    worker thread with a timeout before the pure-CPU parse runs on-loop."""
+    content = await _read_source_async(hc.spec.workflow.resource)
+    return parse_workflow_from_healthcheck(hc, content=content)
+
+
+def parse_remedy_workflow_from_healthcheck(
+    hc: HealthCheck, content: Optional[bytes] = None
+) -> Tuple[Dict[str, Any], Dict[str, str]]:
     """Remedy variant (reference :1002-1125): ``activeDeadlineSeconds``
     defaults from RepeatAfterSec; an existing numeric deadline round-trips
     into ``hc.spec.remedy_workflow.timeout``."""
-    content = b""
-    if hc.spec.remedy_workflow.resource is not None:
-        reader = get_artifact_reader(hc.spec.remedy_workflow.resource.source)
-        content = reader.read()
+    if content is None:
+        content = b""
+        if hc.spec.remedy_workflow.resource is not None:
+            reader = get_artifact_reader(hc.spec.remedy_workflow.resource.source)
+            content = reader.read()
     data = _decode(content, remedy=True)
     labels = _extract_labels(data)
     spec = _validated_spec(data, remedy=True)
@@ -153,3 +200,12 @@ def parse_remedy_workflow_from_healthcheck(hc: HealthCheck) -> Tuple[Dict[str, A
     else:
         hc.spec.remedy_workflow.timeout = timeout
     return spec, labels
+
+
+async def parse_remedy_workflow_from_healthcheck_async(
+    hc: HealthCheck,
+) -> Tuple[Dict[str, Any], Dict[str, str]]:
+    """Event-loop-safe remedy variant (see
+    :func:`parse_workflow_from_healthcheck_async`)."""
+    content = await _read_source_async(hc.spec.remedy_workflow.resource)
+    return parse_remedy_workflow_from_healthcheck(hc, content=content)

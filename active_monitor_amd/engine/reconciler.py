@@ -48,8 +48,8 @@ from ..metrics import (
 from .backoff import IEBTimeoutError, InverseExponentialBackoff, compute_backoff_params
 from .cronx import CronParseError, seconds_until_next
 from .parse import (
-    parse_remedy_workflow_from_healthcheck,
-    parse_workflow_from_healthcheck,
+    parse_remedy_workflow_from_healthcheck_async,
+    parse_workflow_from_healthcheck_async,
 )
 from .rbac import RBACProvisioner
 from .workqueue import WorkQueue
@@ -341,7 +341,7 @@ class HealthCheckReconciler:
 
     async def create_submit_workflow(self, hc: HealthCheck) -> str:
         try:
-            spec, labels = parse_workflow_from_healthcheck(hc)
+            spec, labels = await parse_workflow_from_healthcheck_async(hc)
         except Exception as e:
             await self._event(hc, "Warning", "Error creating or submitting workflow")
             raise e
@@ -365,7 +365,7 @@ class HealthCheckReconciler:
         if hc.spec.remedy_workflow.resource is None:
             raise ValueError("RemedyWorkflow Resource is nil")
         try:
-            spec, labels = parse_remedy_workflow_from_healthcheck(hc)
+            spec, labels = await parse_remedy_workflow_from_healthcheck_async(hc)
         except Exception as e:
             await self._event(hc, "Warning", "Error creating or submitting remedyworkflow")
             raise e

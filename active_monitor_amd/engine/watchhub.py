@@ -13,6 +13,8 @@ hub is purely a wake accelerator ("make the watch non-blocking
 from __future__ import annotations
 
 import asyncio
+import time
+from collections import deque
 from typing import Dict, List, Optional, Tuple
 
 from ..kube.client import KubeClient
@@ -29,6 +31,12 @@ class WorkflowWatchHub:
         # per-key monotonic change counter: lets a watcher detect events that
         # fired between its poll and its wait registration (no lost wakeups)
         self._seq: Dict[Key, int] = {}
+        # keys whose object was DELETED, kept (seq bumped, not popped) so a
+        # late wait_change(since=N) returns immediately and discovers the
+        # NotFound at once instead of sleeping a full IEB interval; pruned
+        # lazily after a grace period — workflow names are generateName-unique
+        # so these entries would otherwise accumulate forever
+        self._deleted: "deque[Tuple[float, Key]]" = deque()
         self._sub = None
         self._task: Optional[asyncio.Task] = None
 
@@ -59,13 +67,22 @@ class WorkflowWatchHub:
             meta = ev["object"].get("metadata") or {}
             key = (meta.get("namespace", ""), meta.get("name", ""))
             waiters = self._waiters.pop(key, ())
+            self._seq[key] = self._seq.get(key, 0) + 1
             if ev["type"] == "DELETED":
-                self._seq.pop(key, None)  # prune; NotFound ends the watchers
-            else:
-                self._seq[key] = self._seq.get(key, 0) + 1
+                self._deleted.append((time.monotonic(), key))
             for fut in waiters:  # wake everyone watching it
                 if not fut.done():
                     fut.set_result(ev["type"])
+            self._prune_deleted()
+
+    _DELETED_TTL = 60.0
+
+    def _prune_deleted(self) -> None:
+        cutoff = time.monotonic() - self._DELETED_TTL
+        while self._deleted and self._deleted[0][0] < cutoff:
+            _, key = self._deleted.popleft()
+            if key not in self._waiters:
+                self._seq.pop(key, None)
 
     async def wait_change(
         self, namespace: str, name: str, timeout: float, since: Optional[int] = None

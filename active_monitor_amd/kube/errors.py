@@ -43,6 +43,14 @@ class InvalidError(ApiError):
     reason = "Invalid"
 
 
+class ExpiredError(ApiError):
+    """410 Gone: a watch resourceVersion older than the server's retained
+    event window (the apiserver's 'too old resource version')."""
+
+    code = 410
+    reason = "Expired"
+
+
 def is_not_found(err: Optional[BaseException]) -> bool:
     return isinstance(err, NotFoundError)
 

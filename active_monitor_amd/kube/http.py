@@ -12,6 +12,7 @@ import asyncio
 import json
 import logging
 import ssl
+import time
 from typing import Any, AsyncIterator, Dict, List, Optional
 
 import aiohttp
@@ -52,8 +53,26 @@ def _error_for(status: int, body: str) -> ApiError:
     return err
 
 
+class _Expired(Exception):
+    """The watch's resourceVersion is too old (HTTP 410 / ERROR event)."""
+
+
 class HttpSubscription:
-    """Streaming watch with automatic reconnect (resourceVersion resume)."""
+    """Streaming watch with automatic reconnect (resourceVersion resume).
+
+    Conformance with client-go's reflector (which the reference inherits via
+    controller-runtime, healthcheck_controller.go:133-137):
+
+    - a 410 Gone response or a watch ``ERROR`` event (code 410 or otherwise)
+      clears the stored resourceVersion, **re-lists** the collection —
+      emitting a synthetic ``ADDED`` per live object so consumers can't miss
+      changes that happened inside the expiry window — and resumes watching
+      from the list's resourceVersion,
+    - ``allowWatchBookmarks`` is requested and BOOKMARK events advance the
+      resume point without being delivered,
+    - frames are read with an incremental buffer, so a watch event larger
+      than aiohttp's 64 KB readline limit cannot wedge the stream.
+    """
 
     def __init__(self, client: "HttpClient", api_version: str, kind: str,
                  namespace: Optional[str]):
@@ -65,6 +84,8 @@ class HttpSubscription:
         self._resource_version: Optional[str] = None
         self._queue: "asyncio.Queue[Optional[Dict[str, Any]]]" = asyncio.Queue()
         self._task = asyncio.ensure_future(self._pump())
+        #: test/telemetry counter: completed re-list recoveries
+        self.relists = 0
 
     async def _pump(self) -> None:
         while not self._closed:
@@ -72,6 +93,21 @@ class HttpSubscription:
                 await self._stream_once()
             except asyncio.CancelledError:
                 return
+            except _Expired:
+                if self._closed:
+                    return
+                log.warning(
+                    "watch resourceVersion expired (%s/%s); re-listing",
+                    self.api_version, self.kind,
+                )
+                self._resource_version = None
+                try:
+                    await self._relist()
+                except asyncio.CancelledError:
+                    return
+                except Exception as e:
+                    log.warning("re-list after watch expiry failed: %s", e)
+                    await asyncio.sleep(1.0)
             except Exception as e:
                 if self._closed:
                     return
@@ -79,8 +115,23 @@ class HttpSubscription:
                             self.api_version, self.kind, e)
                 await asyncio.sleep(1.0)
 
+    async def _relist(self) -> None:
+        """Reflector-style recovery: list the collection, surface every live
+        object as a synthetic ADDED (consumers treat events as level
+        triggers), and resume the watch from the list's resourceVersion."""
+        path = self._client._collection_path(self.api_version, self.kind, self.namespace)
+        out = await self._client._request("GET", path)
+        rv = (out.get("metadata") or {}).get("resourceVersion")
+        if rv:
+            self._resource_version = str(rv)
+        for obj in out.get("items", []):
+            if self._closed:
+                return
+            self._queue.put_nowait({"type": "ADDED", "object": obj})
+        self.relists += 1
+
     async def _stream_once(self) -> None:
-        params = {"watch": "true"}
+        params = {"watch": "true", "allowWatchBookmarks": "true"}
         if self._resource_version:
             params["resourceVersion"] = self._resource_version
         path = self._client._collection_path(self.api_version, self.kind, self.namespace)
@@ -88,21 +139,46 @@ class HttpSubscription:
             self._client.base_url + path, params=params,
             timeout=aiohttp.ClientTimeout(total=None, sock_read=None),
         ) as resp:
+            if resp.status == 410:
+                await resp.text()
+                raise _Expired()
             if resp.status >= 400:
                 raise _error_for(resp.status, await resp.text())
-            async for line in resp.content:
-                if self._closed:
-                    return
-                line = line.strip()
-                if not line:
-                    continue
-                ev = json.loads(line)
-                obj = ev.get("object") or {}
-                rv = (obj.get("metadata") or {}).get("resourceVersion")
-                if rv:
-                    self._resource_version = str(rv)
-                if ev.get("type") in ("ADDED", "MODIFIED", "DELETED"):
-                    self._queue.put_nowait({"type": ev["type"], "object": obj})
+            # incremental framing: newline-delimited JSON without readline's
+            # line-length ceiling (a >64 KB Workflow event must not wedge the
+            # stream in a reconnect-replay loop)
+            buf = bytearray()
+            while True:
+                chunk = await resp.content.readany()
+                if not chunk:
+                    return  # server closed the stream; reconnect with RV
+                buf.extend(chunk)
+                while True:
+                    nl = buf.find(b"\n")
+                    if nl < 0:
+                        break
+                    line = bytes(buf[:nl]).strip()
+                    del buf[: nl + 1]
+                    if not line:
+                        continue
+                    self._handle_event(json.loads(line))
+                    if self._closed:
+                        return
+
+    def _handle_event(self, ev: Dict[str, Any]) -> None:
+        etype = ev.get("type")
+        obj = ev.get("object") or {}
+        if etype == "ERROR":
+            # a watch ERROR carries a metav1.Status; 410 (or anything else —
+            # the stream is unusable either way) triggers re-list recovery
+            raise _Expired()
+        rv = (obj.get("metadata") or {}).get("resourceVersion")
+        if rv:
+            self._resource_version = str(rv)
+        if etype == "BOOKMARK":
+            return  # resume-point advance only, never delivered
+        if etype in ("ADDED", "MODIFIED", "DELETED"):
+            self._queue.put_nowait({"type": etype, "object": obj})
 
     def close(self) -> None:
         self._closed = True
@@ -119,6 +195,34 @@ class HttpSubscription:
         return ev
 
 
+class _TokenBucket:
+    """Client-side request rate limiter (client-go flowcontrol equivalent —
+    the reference inherits controller-runtime's default 20 QPS / burst 30).
+    Waiters are serialized, so throttled requests drain in FIFO order."""
+
+    def __init__(self, qps: float, burst: int):
+        self.qps = qps
+        self.burst = float(burst)
+        self.tokens = float(burst)
+        self.last = time.monotonic()
+        self._lock = asyncio.Lock()
+
+    async def acquire(self) -> None:
+        if self.qps <= 0:
+            return
+        async with self._lock:
+            now = time.monotonic()
+            self.tokens = min(self.burst, self.tokens + (now - self.last) * self.qps)
+            self.last = now
+            if self.tokens >= 1.0:
+                self.tokens -= 1.0
+                return
+            wait = (1.0 - self.tokens) / self.qps
+            self.tokens = 0.0
+            await asyncio.sleep(wait)
+            self.last = time.monotonic()
+
+
 class HttpClient:
     def __init__(
         self,
@@ -127,6 +231,8 @@ class HttpClient:
         verify: bool = True,
         ca_cert: Optional[str] = None,
         registry: Registry = DEFAULT_REGISTRY,
+        qps: float = 20.0,
+        burst: int = 30,
     ):
         self.base_url = base_url.rstrip("/")
         self.token = token
@@ -136,6 +242,8 @@ class HttpClient:
         #: optional (cert_path, key_path) for client-certificate auth
         self.client_cert: tuple = (None, None)
         self._session: Optional[aiohttp.ClientSession] = None
+        # qps<=0 disables throttling (benchmarks measuring the raw wire)
+        self._limiter = _TokenBucket(qps, burst)
 
     async def start(self) -> None:
         headers = {"Content-Type": "application/json"}
@@ -186,6 +294,7 @@ class HttpClient:
                        params: Optional[Dict[str, str]] = None) -> Obj:
         if self._session is None:
             raise RuntimeError("HttpClient.start() must be called before requests")
+        await self._limiter.acquire()
         async with self._session.request(
             method, self.base_url + path,
             json=body if body is not None else None, params=params,

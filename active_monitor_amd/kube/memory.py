@@ -28,7 +28,13 @@ from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
 
 from ..api.types import k8s_now
 from ..utils.fastcopy import deep_copy, snapshot
-from .errors import AlreadyExistsError, ConflictError, InvalidError, NotFoundError
+from .errors import (
+    AlreadyExistsError,
+    ConflictError,
+    ExpiredError,
+    InvalidError,
+    NotFoundError,
+)
 from .registry import DEFAULT_REGISTRY, STATUS_SUBRESOURCE_KINDS, Registry
 
 Obj = Dict[str, Any]
@@ -136,6 +142,16 @@ class MemoryApiServer:
         # counters for observability/benchmarks
         self.op_counts: Dict[str, int] = {"get": 0, "list": 0, "create": 0,
                                           "update": 0, "update_status": 0, "delete": 0}
+        # bounded watch-event history: lets a resumed watch replay exactly the
+        # events after its resourceVersion, and makes 410 Gone REAL — a resume
+        # point older than the window raises ExpiredError, like etcd's
+        # compacted-revision error surfaced by the apiserver. Tests shrink
+        # history_window to force expiry deterministically.
+        self._history: deque = deque()  # (rv:int, ev_type, obj snapshot)
+        self.history_window = 4096
+        # highest rv evicted from history — the compaction horizon; resuming
+        # below it is 410 Gone
+        self._compacted_rv = 0
 
     # -- internals ---------------------------------------------------------
 
@@ -173,8 +189,44 @@ class MemoryApiServer:
 
     def _publish(self, ev_type: str, obj: Obj) -> None:
         event = {"type": ev_type, "object": obj}
+        try:
+            rv = int((obj.get("metadata") or {}).get("resourceVersion", 0))
+        except (TypeError, ValueError):
+            rv = self._rv
+        self._history.append((rv, ev_type, obj))
+        while len(self._history) > self.history_window:
+            self._compacted_rv = self._history.popleft()[0]
         for sub in list(self._subs):
             sub._offer(event)
+
+    def events_since(self, api_version: str, kind: str, namespace: Optional[str],
+                     resource_version: str) -> List[Dict[str, Any]]:
+        """Replay retained events newer than ``resource_version`` for one
+        (apiVersion, kind[, namespace]); raises :class:`ExpiredError` when the
+        resume point predates the retained window (apiserver 410 semantics)."""
+        try:
+            since = int(resource_version)
+        except (TypeError, ValueError):
+            raise ExpiredError(f"resourceVersion {resource_version!r} is invalid")
+        with self._lock:
+            if since < self._compacted_rv:
+                # events in (since, compacted_rv] are gone — the client can't
+                # reconstruct the stream and must re-list
+                raise ExpiredError(
+                    f"too old resource version: {since} ({self._compacted_rv})"
+                )
+            out = []
+            for rv, ev_type, obj in self._history:
+                if rv <= since:
+                    continue
+                if obj.get("apiVersion") != api_version or obj.get("kind") != kind:
+                    continue
+                if namespace is not None and (obj.get("metadata") or {}).get(
+                    "namespace", ""
+                ) != namespace:
+                    continue
+                out.append({"type": ev_type, "object": obj})
+            return out
 
     def _unsubscribe(self, sub: Subscription) -> None:
         with self._lock:
@@ -368,6 +420,9 @@ class MemoryApiServer:
                 return
             del self._objects[key]
             self._unindex_owners(key, obj)
+            # deletion bumps the rv (apiserver semantics) so a watch resuming
+            # from just before the delete replays the DELETED event
+            meta["resourceVersion"] = self._next_rv()
             self._publish("DELETED", snapshot(obj))
             self._cascade_delete(meta.get("uid"))
 
@@ -383,8 +438,14 @@ class MemoryApiServer:
             obj = self._objects.pop(key, None)
             if obj is not None:
                 self._unindex_owners(key, obj)
+                obj["metadata"]["resourceVersion"] = self._next_rv()
                 self._publish("DELETED", snapshot(obj))
                 self._cascade_delete(obj["metadata"].get("uid"))
+
+    def resource_version(self) -> str:
+        """The store's current global resourceVersion (list metadata rv)."""
+        with self._lock:
+            return str(self._rv)
 
     def watch(self, api_version: str, kind: str, namespace: Optional[str] = None) -> Subscription:
         loop = asyncio.get_running_loop()

@@ -16,7 +16,7 @@ from typing import Optional
 
 from aiohttp import web
 
-from .errors import ApiError, InvalidError
+from .errors import ApiError, ExpiredError, InvalidError
 from .memory import MemoryApiServer
 
 log = logging.getLogger("active_monitor_amd.kube.server")
@@ -63,10 +63,16 @@ class ApiServerFrontend:
             self.port = site._server.sockets[0].getsockname()[1]
 
     async def stop(self) -> None:
-        for sub in list(self._live_subs):
-            sub.close()  # end open watch streams so cleanup() is immediate
+        self.kick_watches()
         if self._runner is not None:
             await self._runner.cleanup()
+
+    def kick_watches(self) -> None:
+        """Terminate every open watch stream (server stays up). Clients see a
+        clean end-of-stream and reconnect with their resourceVersion — the
+        fault-injection hook for RV-expiry/resume conformance tests."""
+        for sub in list(self._live_subs):
+            sub.close()  # end open watch streams so cleanup() is immediate
 
     # -- discovery ----------------------------------------------------------
 
@@ -176,9 +182,11 @@ class ApiServerFrontend:
                     api_version, kind, namespace or None,
                     request.query.get("labelSelector"),
                 )
-                return web.json_response(
-                    {"apiVersion": api_version, "kind": kind + "List", "items": items}
-                )
+                return web.json_response({
+                    "apiVersion": api_version, "kind": kind + "List",
+                    "metadata": {"resourceVersion": self.server.resource_version()},
+                    "items": items,
+                })
             if request.method == "GET":
                 return web.json_response(self.server.get(api_version, kind, namespace, name))
             if request.method == "POST":
@@ -213,23 +221,55 @@ class ApiServerFrontend:
             _status_body(ApiError(f"unsupported method {request.method}")), status=405
         )
 
+    @staticmethod
+    def _ev_rv(ev: dict) -> int:
+        try:
+            return int(((ev.get("object") or {}).get("metadata") or {})
+                       .get("resourceVersion", 0))
+        except (TypeError, ValueError):
+            return 0
+
     async def _watch(self, request: web.Request, api_version: str, kind: str,
                      namespace: Optional[str]) -> web.StreamResponse:
-        resp = web.StreamResponse(
-            status=200, headers={"Content-Type": "application/json;stream=watch"}
-        )
-        await resp.prepare(request)
+        """Watch with real resourceVersion semantics: a resume rv replays the
+        retained event history after that rv (410 Gone when it predates the
+        window — the client must re-list, like against a real apiserver); no
+        rv replays current state as ADDED. The live subscription is opened
+        before the replay snapshot is taken and overlap is deduplicated by
+        rv, so no event can fall between replay and stream."""
+        rv_param = request.query.get("resourceVersion")
         sub = self.server.watch(api_version, kind, namespace)
         self._live_subs.append(sub)
         try:
-            # replay current state as ADDED unless resuming from an rv — the
-            # standard list-then-watch contract
-            if not request.query.get("resourceVersion"):
-                for obj in self.server.list(api_version, kind, namespace):
-                    await resp.write(
-                        (json.dumps({"type": "ADDED", "object": obj}) + "\n").encode()
-                    )
+            replay = []
+            last_rv = 0
+            if rv_param:
+                try:
+                    replay = self.server.events_since(api_version, kind, namespace, rv_param)
+                except ExpiredError as e:
+                    sub.close()
+                    self._live_subs.remove(sub)
+                    return web.json_response(_status_body(e), status=410)
+                try:
+                    last_rv = int(rv_param)
+                except ValueError:
+                    last_rv = 0
+            else:
+                last_rv = int(self.server.resource_version())
+                replay = [
+                    {"type": "ADDED", "object": obj}
+                    for obj in self.server.list(api_version, kind, namespace)
+                ]
+            resp = web.StreamResponse(
+                status=200, headers={"Content-Type": "application/json;stream=watch"}
+            )
+            await resp.prepare(request)
+            for ev in replay:
+                last_rv = max(last_rv, self._ev_rv(ev))
+                await resp.write((json.dumps(ev) + "\n").encode())
             async for ev in sub:
+                if self._ev_rv(ev) <= last_rv:
+                    continue  # already covered by the replay snapshot
                 await resp.write((json.dumps(ev) + "\n").encode())
         except (ConnectionResetError, asyncio.CancelledError):
             pass

@@ -13,6 +13,7 @@ available for compatibility tests via ``GetArtifactReaderStrict``.
 """
 from .artifacts import (
     ArtifactReader,
+    ArtifactReadError,
     FileReader,
     InlineReader,
     URLReader,
@@ -21,6 +22,7 @@ from .artifacts import (
 
 __all__ = [
     "ArtifactReader",
+    "ArtifactReadError",
     "FileReader",
     "InlineReader",
     "URLReader",

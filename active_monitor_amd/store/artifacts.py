@@ -47,14 +47,21 @@ class InlineReader(ArtifactReader):
         return self._inline.encode("utf-8")
 
 
+#: hard ceiling on one artifact fetch; the reference's Go http.Client has no
+#: timeout but blocks only its own goroutine — here a bounded read keeps a
+#: hung source from pinning a worker thread forever
+DEFAULT_READ_TIMEOUT = 30.0
+
+
 class URLReader(ArtifactReader):
     """HTTP GET of the workflow YAML; TLS verification on unless
     ``verifyCert: false`` (url.go:20-57, secure by default)."""
 
-    def __init__(self, url: Optional[URLArtifact]):
+    def __init__(self, url: Optional[URLArtifact], timeout: float = DEFAULT_READ_TIMEOUT):
         if url is None:
             raise ArtifactReadError("URLArtifact cannot be empty")
         self._url = url
+        self._timeout = timeout
 
     def read(self) -> bytes:
         log.debug("reading urlArtifact from %s", self._url.path)
@@ -65,7 +72,7 @@ class URLReader(ArtifactReader):
             ctx.check_hostname = False
             ctx.verify_mode = ssl.CERT_NONE
         try:
-            with urllib.request.urlopen(self._url.path, context=ctx) as resp:
+            with urllib.request.urlopen(self._url.path, context=ctx, timeout=self._timeout) as resp:
                 status = getattr(resp, "status", 200)
                 if status != 200:
                     raise ArtifactReadError(f"status code {status}")
@@ -76,6 +83,9 @@ class URLReader(ArtifactReader):
         except urllib.error.URLError as e:
             log.warning("failed to read url %s: %s", self._url.path, e)
             raise ArtifactReadError(str(e)) from e
+        except OSError as e:  # read-side socket timeout / reset
+            log.warning("failed to read url %s: %s", self._url.path, e)
+            raise ArtifactReadError(f"failed to read {self._url.path}: {e}") from e
 
 
 class FileReader(ArtifactReader):
