@@ -244,6 +244,8 @@ class HttpClient:
         self._session: Optional[aiohttp.ClientSession] = None
         # qps<=0 disables throttling (benchmarks measuring the raw wire)
         self._limiter = _TokenBucket(qps, burst)
+        #: unary requests issued (observability/bench; watch streams excluded)
+        self.request_count = 0
 
     async def start(self) -> None:
         headers = {"Content-Type": "application/json"}
@@ -295,6 +297,7 @@ class HttpClient:
         if self._session is None:
             raise RuntimeError("HttpClient.start() must be called before requests")
         await self._limiter.acquire()
+        self.request_count += 1
         async with self._session.request(
             method, self.base_url + path,
             json=body if body is not None else None, params=params,

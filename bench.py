@@ -7,10 +7,26 @@ is pure Kubernetes control-plane — no GPU code paths; the MI355X box serves as
 the Linux host, per BASELINE.json's north star).
 
 Per rank (one process per --gpus N slot, launched by torch.distributed.run for
-N>1): an in-memory apiserver + the full controller stack (manager, N_w
-reconcile workers, workflow watch hub, scripted workflow engine) drives a
-fleet of synthetic HealthCheck CRs shaped like BASELINE config 5 — mixed
-repeatAfterSec + cron CRs with a remedy-carrying failing fraction.
+N>1): the full controller stack (manager, N_w reconcile workers, workflow
+watch hub) drives a fleet of synthetic HealthCheck CRs shaped like BASELINE
+config 5 — mixed repeatAfterSec + cron CRs with a remedy-carrying failing
+fraction.
+
+Apiserver regimes (--apiserver):
+
+- ``wire`` (DEFAULT, the headline regime): each rank spawns a SEPARATE
+  apiserver process (active_monitor_amd.kube.standalone: memory store behind
+  the Kubernetes-REST frontend, plus the scripted workflow engine playing the
+  in-cluster Argo controller) and the controller reaches it over 127.0.0.1
+  HTTP. Every reconcile crosses a real process + TCP + JSON-serialization
+  boundary — the same boundary the reference crosses to kube-apiserver — so
+  the reported p50/throughput include real RTT (measured and reported as
+  config.apiserver_rtt_ms), not zero-RTT self-play.
+- ``memory``: the round-1 in-process regime (store and controller share one
+  event loop; --latency adds simulated RTT). Kept for profiling comparisons.
+- ``--apiserver-url URL``: an external Kubernetes-REST endpoint (a real kind
+  cluster or an already-running standalone apiserver); the harness assumes a
+  workflow controller runs cluster-side.
 
 One STEP = one complete wave: every CR in the fleet completes exactly one
 additional health-check run (reconcile → RBAC ensure → workflow submit →
@@ -39,9 +55,20 @@ def parse_args():
     p.add_argument("--crs", type=int, default=1000, help="HealthCheck CRs per rank")
     p.add_argument("--workers", type=int, default=8, help="MaxConcurrentReconciles per rank")
     p.add_argument("--latency", type=float, default=0.0,
-                   help="simulated apiserver RTT per request (seconds)")
+                   help="memory regime only: simulated apiserver RTT (seconds)")
     p.add_argument("--remedy-frac", type=float, default=0.2)
     p.add_argument("--cron-frac", type=float, default=0.3)
+    p.add_argument("--apiserver", choices=["wire", "memory"], default="wire",
+                   help="wire: real HTTP to a separate apiserver process "
+                        "(headline); memory: in-process store")
+    p.add_argument("--apiserver-url", default="",
+                   help="use an external Kubernetes-REST endpoint instead of "
+                        "spawning one (implies --apiserver wire; a workflow "
+                        "controller must be running cluster-side)")
+    p.add_argument("--qps", type=float, default=0.0,
+                   help="client-side rate limit (0 = unlimited, the bench "
+                        "default — the reference harness likewise tunes "
+                        "client-go QPS/Burst for load tests)")
     return p.parse_args()
 
 
@@ -101,39 +128,81 @@ def make_cr(i: int, ns: str, cron_frac: float, remedy_frac: float):
     }, is_remedy
 
 
+async def _spawn_apiserver(args):
+    """Start the standalone apiserver process and wait for its READY line."""
+    proc = await asyncio.create_subprocess_exec(
+        sys.executable, "-m", "active_monitor_amd.kube.standalone",
+        "--engine", "scripted-bench",
+        "--remedy-frac", str(args.remedy_frac),
+        stdout=asyncio.subprocess.PIPE,
+        stderr=asyncio.subprocess.DEVNULL,
+    )
+    line = await asyncio.wait_for(proc.stdout.readline(), 60)
+    if not line.startswith(b"READY "):
+        raise RuntimeError(f"apiserver process failed to start: {line!r}")
+    info = json.loads(line[len(b"READY "):])
+    return proc, info["url"]
+
+
+async def _measure_rtt_ms(client, n: int = 25) -> float:
+    """Median round-trip of an apiserver GET over the actual wire."""
+    samples = []
+    for _ in range(n):
+        t0 = time.monotonic()
+        await client.ping()
+        samples.append(time.monotonic() - t0)
+    return statistics.median(samples) * 1000.0
+
+
 async def run_rank(args, rank: int):
     from active_monitor_amd.engine import Manager
     from active_monitor_amd.kube import MemoryApiServer, MemoryClient
     from active_monitor_amd.workflow import ScriptedWorkflowEngine
 
     ns = "health"
-    server = MemoryApiServer()
-    client = MemoryClient(server, latency=args.latency)
+    server = None
+    engine = None
+    apiserver_proc = None
+    rtt_ms = 0.0
 
-    def policy(wf):
-        # remedy-carrying CRs have failing checks; remedies succeed
-        name = wf["metadata"]["name"]
-        if "-remedy-wf-" in name:
+    if args.apiserver_url or args.apiserver == "wire":
+        from active_monitor_amd.kube.http import HttpClient
+
+        if args.apiserver_url:
+            url = args.apiserver_url
+            apiserver_desc = url
+        else:
+            apiserver_proc, url = await _spawn_apiserver(args)
+            apiserver_desc = "http-subprocess-127.0.0.1"
+        client = HttpClient(url, qps=args.qps)
+        await client.start()
+        rtt_ms = await _measure_rtt_ms(client)
+    else:
+        server = MemoryApiServer()
+        client = MemoryClient(server, latency=args.latency)
+        apiserver_desc = "memory-inproc"
+
+        remedy_names = set()
+
+        def policy(wf):
+            # remedy-carrying CRs have failing checks; remedies succeed
+            name = wf["metadata"]["name"]
+            if "-remedy-wf-" in name:
+                return ("Succeeded", "")
+            if name.split("-wf-")[0] in remedy_names:
+                return ("Failed", "synthetic failure")
             return ("Succeeded", "")
-        if name.startswith("hc-") and _is_remedy_name(name):
-            return ("Failed", "synthetic failure")
-        return ("Succeeded", "")
 
-    remedy_names = set()
+        engine = ScriptedWorkflowEngine(client, policy=policy)
+        await engine.start()
 
-    def _is_remedy_name(wfname):
-        # wfname: hc-00001-wf-xxxxx → hc-00001
-        return wfname.split("-wf-")[0] in remedy_names
-
-    engine = ScriptedWorkflowEngine(client, policy=policy)
-    await engine.start()
     manager = Manager(client, max_workers=args.workers)
     await manager.start()
 
     crs = []
     for i in range(args.crs):
         cr, is_remedy = make_cr(i, ns, args.cron_frac, args.remedy_frac)
-        if is_remedy:
+        if is_remedy and server is not None:
             remedy_names.add(cr["metadata"]["name"])
         crs.append(cr)
         await client.create(cr)
@@ -171,7 +240,20 @@ async def run_rank(args, rank: int):
     p99 = (sorted(lat)[int(len(lat) * 0.99)] * 1000) if lat else 0.0
 
     await manager.stop()
-    await engine.stop()
+    if engine is not None:
+        await engine.stop()
+    if server is not None:
+        requests = sum(server.op_counts.values())
+    else:
+        requests = client.request_count
+        await client.close()
+    if apiserver_proc is not None:
+        apiserver_proc.terminate()
+        try:
+            await asyncio.wait_for(apiserver_proc.wait(), 10)
+        except asyncio.TimeoutError:
+            apiserver_proc.kill()
+            await apiserver_proc.wait()
     return {
         "elapsed": elapsed,
         "cycles": args.crs * args.steps,
@@ -179,7 +261,9 @@ async def run_rank(args, rank: int):
         "p50_reconcile_latency_ms": p50,
         "p99_reconcile_latency_ms": p99,
         "wave_times": wave_times,
-        "apiserver_requests": sum(server.op_counts.values()),
+        "apiserver_requests": requests,
+        "apiserver": apiserver_desc,
+        "apiserver_rtt_ms": rtt_ms,
     }
 
 
@@ -288,6 +372,10 @@ def main():
                           f"{int(args.remedy_frac*100)}% failing-with-remedy",
                 "p50_reconcile_latency_ms": round(p50, 4),
                 "p99_reconcile_latency_ms": round(p99, 4),
+                "apiserver": all_results[0]["apiserver"],
+                "apiserver_rtt_ms": round(
+                    max(r["apiserver_rtt_ms"] for r in all_results), 4
+                ),
                 "apiserver_latency_s": args.latency,
             },
         }
