@@ -86,6 +86,15 @@ class Manager:
         # pure inverse-exponential polling (same semantics, higher latency)
         self.enable_wf_hub = enable_wf_hub
         self.wf_hub = None
+        # informer cache: the watch stream already delivers every HealthCheck
+        # object; serving reconciler reads from it (controller-runtime cached
+        # client, healthcheck_controller.go:133) saves ~2.5 apiserver GETs per
+        # cycle on the wire. Events update the cache BEFORE enqueueing, so a
+        # dequeued key's cache entry is at least as new as the event that
+        # queued it. Only this shard's keys are cached.
+        self.hc_cache: dict = {}
+        self._cache_synced = False
+        self.enable_hc_cache = True
 
     # -- lifecycle ----------------------------------------------------------
 
@@ -173,6 +182,14 @@ class Manager:
 
         return zlib.crc32(name.encode()) % self.shard_count == self.shard_index
 
+    def _cache_lookup(self, namespace: str, name: str):
+        """Reconciler-facing cache read; see HealthCheckReconciler.hc_lookup."""
+        from .reconciler import CACHE_MISS
+
+        if not self._cache_synced:
+            return CACHE_MISS
+        return self.hc_cache.get((namespace, name))
+
     async def _informer(self) -> None:
         self._sub = self.client.watch(API_VERSION, HC_KIND, self.namespace)
         # initial list AFTER subscribing so no event can slip between the two
@@ -180,12 +197,22 @@ class Manager:
             meta = obj.get("metadata") or {}
             name = meta.get("name", "")
             if self._owns(name):
+                self.hc_cache[(meta.get("namespace", ""), name)] = obj
                 await self.queue.add((meta.get("namespace", ""), name))
+        if self.enable_hc_cache:
+            self._cache_synced = True
+            self.reconciler.hc_lookup = self._cache_lookup
         async for ev in self._sub:
-            meta = ev["object"].get("metadata") or {}
+            obj = ev["object"]
+            meta = obj.get("metadata") or {}
             name = meta.get("name", "")
             if self._owns(name):
-                await self.queue.add((meta.get("namespace", ""), name))
+                key = (meta.get("namespace", ""), name)
+                if ev["type"] == "DELETED":
+                    self.hc_cache.pop(key, None)
+                else:
+                    self.hc_cache[key] = obj
+                await self.queue.add(key)
 
     # -- workers ------------------------------------------------------------
 

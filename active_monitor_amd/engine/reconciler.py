@@ -63,6 +63,10 @@ REMEDY = "remedy"
 HEALTHCHECK = "healthCheck"
 TIMER_FLAG = "timer"
 
+#: sentinel returned by an informer-cache lookup that cannot answer (cache
+#: not yet synced); distinct from None, which means "synced and not found"
+CACHE_MISS = object()
+
 
 @dataclass
 class ReconcileResult:
@@ -101,6 +105,12 @@ class HealthCheckReconciler:
         # optional WorkflowWatchHub (set by the Manager): event-driven wakeups
         # for workflow completion; None falls back to pure IEB polling
         self.wf_hub = None
+        # optional informer-cache lookup (set by the Manager): serves
+        # HealthCheck reads from the watch-fed cache like controller-runtime's
+        # cached client (healthcheck_controller.go:133 reads through the
+        # informer cache, not the wire). Returns a dict, None (synced and
+        # absent ⇒ NotFound) or CACHE_MISS (not synced / not covered).
+        self.hc_lookup = None
         self.repeat_timers_by_name: Dict[str, RepeatTimer] = {}
         self._watch_tasks: Dict[str, Set[asyncio.Task]] = {}
         # observability for benchmarks/tests
@@ -206,14 +216,30 @@ class HealthCheckReconciler:
     # Reconcile entry (reference :170-223)
     # ------------------------------------------------------------------
 
+    async def _get_hc(self, namespace: str, name: str) -> Dict[str, Any]:
+        """Read a HealthCheck through the informer cache when one is synced,
+        else over the wire. Cache hits return a shallow top-level copy so the
+        caller may replace top-level keys (the status-update path does)
+        without corrupting the shared cached object."""
+        lk = self.hc_lookup
+        if lk is not None:
+            hit = lk(namespace, name)
+            if hit is None:
+                raise NotFoundError(
+                    f'healthchecks.activemonitor.keikoproj.io "{name}" not found'
+                )
+            if hit is not CACHE_MISS:
+                return dict(hit)
+        return await self.client.get(API_VERSION, HC_KIND, namespace, name,
+                                     snapshot_read=True)
+
     async def reconcile(
         self, namespace: str, name: str, flags: Optional[Set[str]] = None
     ) -> ReconcileResult:
         flags = flags or set()
         self.reconcile_count += 1
         try:
-            obj = await self.client.get(API_VERSION, HC_KIND, namespace, name,
-                                        snapshot_read=True)
+            obj = await self._get_hc(namespace, name)
         except NotFoundError:
             # CR deleted: stop the repeat timer so self-scheduling halts
             # (:180-184); in-flight watches are cancelled proactively (the
@@ -391,11 +417,31 @@ class HealthCheckReconciler:
 
     _POLL_RETRIES = 3
 
-    async def _poll_workflow(self, namespace: str, name: str) -> Optional[Dict[str, Any]]:
+    async def _poll_workflow(
+        self, namespace: str, name: str, via_cache: bool = False
+    ) -> Optional[Dict[str, Any]]:
         """One status poll. NotFound propagates (ends the watch, :618-622);
         transient apiserver errors are retried briefly and then treated as a
         missed poll — the reference aborts the whole watch on any error,
-        stalling the CR until an external reconcile."""
+        stalling the CR until an external reconcile.
+
+        ``via_cache=True`` reads from the watch hub's last-event cache (the
+        watch payload already carried the full Workflow) instead of a GET
+        round-trip — the informer-cache read discipline of controller-runtime.
+        A cache miss means no event reached the hub yet, i.e. the freshly
+        submitted workflow has no status — reported as such without touching
+        the wire. The IEB fallback polls (hub wait timed out) always go
+        direct, so a lost watch event costs one poll interval, never
+        correctness."""
+        if via_cache and self.wf_hub is not None:
+            hit = self.wf_hub.cached(namespace, name)
+            if hit is not None:
+                ev_type, obj = hit
+                if ev_type == "DELETED":
+                    raise NotFoundError(f'workflow "{name}" not found (deleted)')
+                status = obj.get("status")
+                return status if isinstance(status, dict) else None
+            return None
         last: Optional[BaseException] = None
         for attempt in range(self._POLL_RETRIES):
             try:
@@ -415,20 +461,23 @@ class HealthCheckReconciler:
 
     async def _wait_next_poll(
         self, ieb: InverseExponentialBackoff, namespace: str, name: str, since: Optional[int]
-    ) -> None:
+    ) -> bool:
         """Wait until the next poll is due. With a watch hub the wait ends the
         moment the workflow changes (ms-scale completion detection); without
         one this is exactly the reference's IEB sleep. Raises IEBTimeoutError
         past the deadline either way — the synthesized-failure semantics are
-        identical."""
+        identical. Returns True when a watch event (not a timeout) ended the
+        wait, i.e. the hub cache holds the fresh object and the next poll may
+        read it without a wire round-trip."""
         if self.wf_hub is None:
             await ieb.next()
-            return
+            return False
         ieb.check_deadline()
         interval = ieb.peek_interval()
         ieb.decay()
-        await self.wf_hub.wait_change(namespace, name, interval, since=since)
+        changed = await self.wf_hub.wait_change(namespace, name, interval, since=since)
         ieb.check_deadline()
+        return changed is not None
 
     async def watch_workflow_reschedule(
         self, wf_namespace: str, wf_name: str, hc: HealthCheck
@@ -449,12 +498,16 @@ class HealthCheckReconciler:
             # constructor error on its first iteration and synthesizes Failed
             timed_out = True
 
+        # the first poll may read the hub cache (a just-submitted workflow has
+        # no status until an event lands); timed-out waits force direct polls
+        via_cache = self.wf_hub is not None
         while True:
             # pre-poll change counter: any event landing between this poll and
             # the wait below is detected immediately (no lost wakeups)
             seq = self.wf_hub.seq(wf_namespace, wf_name) if self.wf_hub else None
             try:
-                status = await self._poll_workflow(wf_namespace, wf_name)
+                status = await self._poll_workflow(wf_namespace, wf_name,
+                                                   via_cache=via_cache)
             except NotFoundError:
                 # parent healthcheck likely deleted; don't reschedule (:618-622)
                 await self._event(
@@ -527,7 +580,7 @@ class HealthCheckReconciler:
                     break
             # not terminal yet: wait for the next poll (hub-accelerated)
             try:
-                await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
+                via_cache = await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
             except IEBTimeoutError:
                 timed_out = True
 
@@ -573,8 +626,7 @@ class HealthCheckReconciler:
         closing a duplicate-submission race the reference leaves open by
         updating first (:734) and arming after (:746)."""
         try:
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
-                                          snapshot_read=True)
+            fresh = await self._get_hc(hc.namespace, hc.name)
         except NotFoundError:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):
@@ -636,10 +688,12 @@ class HealthCheckReconciler:
         except ValueError:
             timed_out = True
 
+        via_cache = self.wf_hub is not None
         while True:
             seq = self.wf_hub.seq(wf_namespace, wf_name) if self.wf_hub else None
             try:
-                status = await self._poll_workflow(wf_namespace, wf_name)
+                status = await self._poll_workflow(wf_namespace, wf_name,
+                                                   via_cache=via_cache)
             except NotFoundError:
                 return
             if timed_out:
@@ -689,14 +743,13 @@ class HealthCheckReconciler:
                     MonitorFinishedTime.labels(hc.name, REMEDY).set(int(now_unix))
                     break
             try:
-                await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
+                via_cache = await self._wait_next_poll(ieb, wf_namespace, wf_name, seq)
             except IEBTimeoutError:
                 timed_out = True
 
         # persist remedy status promptly (reference :856-871)
         try:
-            fresh = await self.client.get(API_VERSION, HC_KIND, hc.namespace, hc.name,
-                                          snapshot_read=True)
+            fresh = await self._get_hc(hc.namespace, hc.name)
         except NotFoundError:
             return
         if (fresh.get("metadata") or {}).get("deletionTimestamp"):

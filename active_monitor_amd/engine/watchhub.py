@@ -37,11 +37,22 @@ class WorkflowWatchHub:
         # lazily after a grace period — workflow names are generateName-unique
         # so these entries would otherwise accumulate forever
         self._deleted: "deque[Tuple[float, Key]]" = deque()
+        # last event per key: the watch payload already carries the full
+        # Workflow object, so completion polls can read status from here
+        # instead of a GET round-trip (controller-runtime informer-cache
+        # shape). Entries are read-only shared snapshots.
+        self._last: Dict[Key, Tuple[str, dict]] = {}
         self._sub = None
         self._task: Optional[asyncio.Task] = None
 
     def seq(self, namespace: str, name: str) -> int:
         return self._seq.get((namespace, name), 0)
+
+    def cached(self, namespace: str, name: str) -> Optional[Tuple[str, dict]]:
+        """Last (event_type, object) seen for the workflow, or None if no
+        event reached the hub yet. The object is a shared read-only snapshot;
+        an entry with event_type DELETED means the object is gone."""
+        return self._last.get((namespace, name))
 
     async def start(self) -> None:
         self._sub = self.client.watch(WF_API_VERSION, WF_KIND, self.namespace)
@@ -68,6 +79,7 @@ class WorkflowWatchHub:
             key = (meta.get("namespace", ""), meta.get("name", ""))
             waiters = self._waiters.pop(key, ())
             self._seq[key] = self._seq.get(key, 0) + 1
+            self._last[key] = (ev["type"], ev["object"])
             if ev["type"] == "DELETED":
                 self._deleted.append((time.monotonic(), key))
             for fut in waiters:  # wake everyone watching it
@@ -83,6 +95,7 @@ class WorkflowWatchHub:
             _, key = self._deleted.popleft()
             if key not in self._waiters:
                 self._seq.pop(key, None)
+                self._last.pop(key, None)
 
     async def wait_change(
         self, namespace: str, name: str, timeout: float, since: Optional[int] = None

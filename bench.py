@@ -376,6 +376,13 @@ def main():
                 "apiserver_rtt_ms": round(
                     max(r["apiserver_rtt_ms"] for r in all_results), 4
                 ),
+                # wire efficiency: unary apiserver requests per completed
+                # cycle (includes warmup/settle traffic; informer-cached
+                # reads and event aggregation keep this low)
+                "apiserver_requests_per_cycle": round(
+                    sum(r["apiserver_requests"] for r in all_results)
+                    / max(1, sum(r["cycles"] for r in all_results)), 2
+                ),
                 "apiserver_latency_s": args.latency,
             },
         }
