@@ -39,6 +39,20 @@ def build_parser() -> argparse.ArgumentParser:
     )
     p.add_argument("--metrics-bind-address", default=":8443",
                    help="metrics endpoint address ('0' disables)")
+    p.add_argument("--metrics-secure", default=True,
+                   action=argparse.BooleanOptionalAction,
+                   help="serve /metrics over HTTPS with bearer-token authn "
+                        "(reference default true, cmd/main.go:139); "
+                        "--no-metrics-secure serves plain HTTP")
+    p.add_argument("--metrics-cert", default=None,
+                   help="TLS certificate for the metrics endpoint "
+                        "(default: self-signed generated at startup)")
+    p.add_argument("--metrics-key", default=None,
+                   help="TLS key for the metrics endpoint")
+    p.add_argument("--metrics-auth-token-file", default=None,
+                   help="file holding the bearer token scrapers must present "
+                        "(default: a random token generated at startup; its "
+                        "path is logged)")
     p.add_argument("--health-probe-bind-address", default=":8081",
                    help="healthz/readyz endpoint address ('0' disables)")
     p.add_argument("--leader-elect", action="store_true",
@@ -138,15 +152,28 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
             await frontend.start()
             log.info("serving Kubernetes REST API at %s", frontend.url)
 
+    metrics_addr = parse_bind_address(args.metrics_bind_address)
+    metrics_security = None
+    if metrics_addr is not None and getattr(args, "metrics_secure", False):
+        from ..engine.endpoints import build_metrics_security
+
+        metrics_security = build_metrics_security(
+            True,
+            cert=getattr(args, "metrics_cert", None),
+            key=getattr(args, "metrics_key", None),
+            token_file=getattr(args, "metrics_auth_token_file", None),
+        )
+
     manager = Manager(
         client,
         max_workers=args.max_workers,
         namespace=args.namespace,
-        metrics_addr=parse_bind_address(args.metrics_bind_address),
+        metrics_addr=metrics_addr,
         health_addr=parse_bind_address(args.health_probe_bind_address),
         leader_elect=args.leader_elect,
         shard_index=args.shard_index,
         shard_count=args.shard_count,
+        metrics_security=metrics_security,
     )
 
     stop = stop_event if stop_event is not None else asyncio.Event()

@@ -43,6 +43,7 @@ class Manager:
         health_addr: Optional[Tuple[str, int]] = None,
         leader_elect: bool = False,
         leader_identity: str = "",
+        metrics_security=None,
         record_latencies: bool = True,
         shard_index: int = 0,
         shard_count: int = 1,
@@ -58,6 +59,9 @@ class Manager:
         )
         self.metrics_addr = metrics_addr
         self.health_addr = health_addr
+        # optional endpoints.MetricsSecurity: TLS + bearer authn on /metrics
+        # (reference default is the secure path, cmd/main.go:74-85,139)
+        self.metrics_security = metrics_security
         self.leader_elect = leader_elect
         self.leader_identity = leader_identity
         self._tasks: List[asyncio.Task] = []
@@ -118,7 +122,8 @@ class Manager:
             from .endpoints import serve_endpoints
 
             self._servers = await serve_endpoints(
-                self, health=self.health_addr, metrics=self.metrics_addr
+                self, health=self.health_addr, metrics=self.metrics_addr,
+                metrics_security=self.metrics_security,
             )
 
         if self.enable_wf_hub:

@@ -26,6 +26,9 @@ def test_run_memory_backend_startup_and_shutdown(run):
             "--max-workers", "3",
             "--metrics-bind-address", "127.0.0.1:18181",
             "--health-probe-bind-address", "127.0.0.1:18182",
+            # plain-HTTP scrape: this test covers lifecycle; the secure
+            # default path is covered by tests/unit/test_metrics_security.py
+            "--no-metrics-secure",
         ])
         stop = asyncio.Event()
         task = asyncio.ensure_future(cli_run(args, stop))
