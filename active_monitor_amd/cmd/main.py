@@ -79,6 +79,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="this controller's shard (keyspace split by CR-name hash)")
     p.add_argument("--shard-count", type=int, default=1,
                    help="total cooperating controller shards")
+    p.add_argument("--shard-ha", action="store_true",
+                   help="lease-per-shard failure takeover: live shards adopt "
+                        "a dead shard's keys; a restarted shard reclaims its "
+                        "keys via the lease preferredHolder handshake")
     p.add_argument("--zap-log-level", default="info",
                    help="log level (debug/info/warn/error)")
     p.add_argument("--log-format", choices=["console", "json"], default="console",
@@ -173,6 +177,7 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         leader_elect=args.leader_elect,
         shard_index=args.shard_index,
         shard_count=args.shard_count,
+        shard_ha=getattr(args, "shard_ha", False),
         metrics_security=metrics_security,
     )
 

@@ -48,6 +48,9 @@ class Manager:
         shard_index: int = 0,
         shard_count: int = 1,
         enable_wf_hub: bool = True,
+        shard_ha: bool = False,
+        shard_lease_duration: float = 15.0,
+        shard_renew_interval: float = 5.0,
     ):
         self.client = client
         self.max_workers = max_workers
@@ -86,6 +89,13 @@ class Manager:
             raise ValueError("shard_index must be in [0, shard_count)")
         self.shard_index = shard_index
         self.shard_count = shard_count
+        # shard failure takeover (engine/shards.py): lease-per-shard with
+        # adoption of dead shards and preferredHolder reclaim. Off by default
+        # (static sharding, the round-1 behavior) — enable with --shard-ha.
+        self.shard_ha = shard_ha and shard_count > 1
+        self.shard_lease_duration = shard_lease_duration
+        self.shard_renew_interval = shard_renew_interval
+        self.coordinator = None
         # event-driven completion wakeups; False falls back to the reference's
         # pure inverse-exponential polling (same semantics, higher latency)
         self.enable_wf_hub = enable_wf_hub
@@ -118,6 +128,25 @@ class Manager:
             self._tasks.append(renew)
             self.elector = elector
 
+        if self.shard_ha:
+            from .shards import ShardCoordinator
+
+            self.coordinator = ShardCoordinator(
+                self.client,
+                namespace=self.namespace or "default",
+                shard_index=self.shard_index,
+                shard_count=self.shard_count,
+                identity=self.leader_identity or f"shard-{self.shard_index}-{id(self):x}",
+                lease_duration=self.shard_lease_duration,
+                renew_interval=self.shard_renew_interval,
+                on_adopt=self._adopt_shard,
+                on_drop=self._drop_shard,
+                on_home_lost=self._home_shard_lost,
+            )
+            # blocks until the home shard's lease is held (a live previous
+            # holder is asked to hand it over via preferredHolder)
+            await self.coordinator.start()
+
         if self.health_addr is not None or self.metrics_addr is not None:
             from .endpoints import serve_endpoints
 
@@ -147,6 +176,8 @@ class Manager:
 
     async def stop(self) -> None:
         self._stopped = True
+        if self.coordinator is not None:
+            await self.coordinator.stop()
         if self._sub is not None:
             self._sub.close()
         if getattr(self, "wf_hub", None) is not None:
@@ -161,6 +192,51 @@ class Manager:
         await asyncio.gather(*self._tasks, return_exceptions=True)
         for srv in self._servers:
             srv.close()
+
+    # -- shard-HA reactions (engine/shards.py callbacks) --------------------
+
+    async def _adopt_shard(self, shard: int) -> None:
+        """An orphaned shard is ours now: surface its CRs (cache + queue) so
+        normal reconciles re-arm their timers and resume their schedules —
+        the restart-resume path, applied to a subset of the keyspace."""
+        from .shards import shard_of
+
+        try:
+            objs = await self.client.list(API_VERSION, HC_KIND, self.namespace)
+        except Exception as e:
+            log.error("adopt shard %d: list failed: %s", shard, e)
+            return
+        n = 0
+        for obj in objs:
+            meta = obj.get("metadata") or {}
+            name = meta.get("name", "")
+            if shard_of(name, self.shard_count) != shard:
+                continue
+            key = (meta.get("namespace", ""), name)
+            self.hc_cache[key] = obj
+            await self.queue.add(key)
+            n += 1
+        log.warning("adopted shard %d: %d healthchecks enqueued", shard, n)
+
+    async def _drop_shard(self, shard: int) -> None:
+        """The rightful owner reclaimed the shard: stop its timers/watches and
+        forget its CRs so exactly one process drives each key."""
+        from .shards import shard_of
+
+        dropped = 0
+        for (ns, name) in list(self.hc_cache):
+            if shard_of(name, self.shard_count) != shard:
+                continue
+            self.hc_cache.pop((ns, name), None)
+            self.reconciler._stop_timer(name, ns)
+            self.reconciler._cancel_watches((ns, name))
+            dropped += 1
+        log.info("dropped shard %d: released %d healthchecks", shard, dropped)
+
+    def _home_shard_lost(self) -> None:
+        self.fatal_reason = f"home shard {self.shard_index} lease lost"
+        log.error("%s — shutting down", self.fatal_reason)
+        self.fatal.set()
 
     def _renew_done(self, task: "asyncio.Task") -> None:
         """A finished renew loop means the lease is gone: a deposed replica
@@ -182,10 +258,12 @@ class Manager:
     def _owns(self, name: str) -> bool:
         if self.shard_count <= 1:
             return True
-        # stable, process-independent hash (builtin hash() is salted)
-        import zlib
+        from .shards import shard_of
 
-        return zlib.crc32(name.encode()) % self.shard_count == self.shard_index
+        shard = shard_of(name, self.shard_count)
+        if self.coordinator is not None:
+            return shard in self.coordinator.owned
+        return shard == self.shard_index
 
     def _cache_lookup(self, namespace: str, name: str):
         """Reconciler-facing cache read; see HealthCheckReconciler.hc_lookup."""
