@@ -83,6 +83,10 @@ def build_parser() -> argparse.ArgumentParser:
                    help="lease-per-shard failure takeover: live shards adopt "
                         "a dead shard's keys; a restarted shard reclaims its "
                         "keys via the lease preferredHolder handshake")
+    p.add_argument("--shard-lease-duration", type=float, default=15.0,
+                   help="shard-ha lease TTL seconds (takeover latency bound)")
+    p.add_argument("--shard-renew-interval", type=float, default=5.0,
+                   help="shard-ha lease renew/scan cadence seconds")
     p.add_argument("--zap-log-level", default="info",
                    help="log level (debug/info/warn/error)")
     p.add_argument("--log-format", choices=["console", "json"], default="console",
@@ -178,6 +182,8 @@ async def _run(args, stop_event: Optional[asyncio.Event] = None) -> int:
         shard_index=args.shard_index,
         shard_count=args.shard_count,
         shard_ha=getattr(args, "shard_ha", False),
+        shard_lease_duration=getattr(args, "shard_lease_duration", 15.0),
+        shard_renew_interval=getattr(args, "shard_renew_interval", 5.0),
         metrics_security=metrics_security,
     )
 

@@ -38,11 +38,29 @@ async def main_async(args) -> dict:
     from active_monitor_amd.kube import MemoryApiServer, MemoryClient
     from active_monitor_amd.workflow import ScriptedWorkflowEngine
 
-    server = MemoryApiServer()
-    client = MemoryClient(server)
-    engine = ScriptedWorkflowEngine(client, policy=lambda wf: ("Succeeded", ""),
-                                    ttl_seconds=args.ttl)
-    await engine.start()
+    engine = None
+    apiserver_proc = None
+    if args.apiserver == "wire":
+        # real-wire regime (bench.py's default): separate apiserver process,
+        # controller over 127.0.0.1 HTTP, the engine in the apiserver process
+        from active_monitor_amd.kube.http import HttpClient
+
+        apiserver_proc = await asyncio.create_subprocess_exec(
+            sys.executable, "-m", "active_monitor_amd.kube.standalone",
+            "--engine", "scripted-bench", "--remedy-frac", "0",
+            stdout=asyncio.subprocess.PIPE, stderr=asyncio.subprocess.DEVNULL,
+        )
+        line = await asyncio.wait_for(apiserver_proc.stdout.readline(), 60)
+        url = json.loads(line[len(b"READY "):])["url"]
+        server = None
+        client = HttpClient(url, qps=0)
+        await client.start()
+    else:
+        server = MemoryApiServer()
+        client = MemoryClient(server)
+        engine = ScriptedWorkflowEngine(client, policy=lambda wf: ("Succeeded", ""),
+                                        ttl_seconds=args.ttl)
+        await engine.start()
     manager = Manager(client, max_workers=args.workers)
     await manager.start()
 
@@ -87,7 +105,7 @@ async def main_async(args) -> dict:
             "rss_mb": round(rss_mb(), 1),
             "active_watches": rec.active_watches(),
             "queue": len(manager.queue),
-            "objects": len(server),
+            "objects": len(server) if server is not None else None,
         })
     elapsed = time.monotonic() - t0
     total = rec.completed_runs - runs_start
@@ -102,11 +120,17 @@ async def main_async(args) -> dict:
         "p50_reconcile_ms": round(statistics.median(lat) * 1000, 4) if lat else None,
         "rss_start_mb": round(rss_start, 1),
         "rss_end_mb": round(rss_mb(), 1),
-        "store_objects_end": len(server),
+        "store_objects_end": len(server) if server is not None else None,
+        "apiserver": args.apiserver,
         "samples": samples,
     }
     await manager.stop()
-    await engine.stop()
+    if engine is not None:
+        await engine.stop()
+    if apiserver_proc is not None:
+        await client.close()
+        apiserver_proc.terminate()
+        await apiserver_proc.wait()
     return result
 
 
@@ -121,6 +145,10 @@ def main() -> int:
     ap.add_argument("--duration", type=float, default=120.0)
     ap.add_argument("--ttl", type=float, default=30.0,
                     help="completed-workflow TTL (Argo ttlStrategy equivalent)")
+    ap.add_argument("--apiserver", choices=["wire", "memory"], default="wire",
+                    help="wire: separate apiserver process over 127.0.0.1 "
+                         "HTTP (the bench.py headline regime); memory: "
+                         "in-process store")
     args = ap.parse_args()
     print(json.dumps(asyncio.run(main_async(args))))
     return 0
