@@ -122,6 +122,10 @@ async def main_async(args) -> dict:
         "rss_end_mb": round(rss_mb(), 1),
         "store_objects_end": len(server) if server is not None else None,
         "apiserver": args.apiserver,
+        "apiserver_requests_per_cycle": (
+            round(client.request_count / max(1, rec.completed_runs), 2)
+            if apiserver_proc is not None else None
+        ),
         "samples": samples,
     }
     await manager.stop()
