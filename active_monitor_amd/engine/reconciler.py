@@ -517,6 +517,8 @@ class HealthCheckReconciler:
                     "indicate that either the healthcheck was removed or the Workflow "
                     "was GC'd before active-monitor could obtain the status",
                 )
+                if self.wf_hub is not None:
+                    self.wf_hub.forget(wf_namespace, wf_name)
                 return
             if timed_out:
                 status = {"phase": FAIL_STR, "message": FAIL_STR}
@@ -587,6 +589,11 @@ class HealthCheckReconciler:
         try:
             await self._finish_and_reschedule(hc, wf_namespace, wf_name, repeat_after_sec)
         finally:
+            # this watch is done with the workflow forever (names are unique):
+            # release its hub cache/seq entry now rather than waiting for the
+            # server-side TTL delete (RSS leak at fleet rates otherwise)
+            if self.wf_hub is not None:
+                self.wf_hub.forget(wf_namespace, wf_name)
             self.completed_runs += 1
 
     async def _maybe_run_remedy(self, hc: HealthCheck, now_unix: float) -> None:
@@ -695,6 +702,8 @@ class HealthCheckReconciler:
                 status = await self._poll_workflow(wf_namespace, wf_name,
                                                    via_cache=via_cache)
             except NotFoundError:
+                if self.wf_hub is not None:
+                    self.wf_hub.forget(wf_namespace, wf_name)
                 return
             if timed_out:
                 status = {"phase": FAIL_STR, "message": FAIL_STR}
@@ -747,6 +756,8 @@ class HealthCheckReconciler:
             except IEBTimeoutError:
                 timed_out = True
 
+        if self.wf_hub is not None:  # done with this workflow forever
+            self.wf_hub.forget(wf_namespace, wf_name)
         # persist remedy status promptly (reference :856-871)
         try:
             fresh = await self._get_hc(hc.namespace, hc.name)

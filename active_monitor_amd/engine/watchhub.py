@@ -54,6 +54,17 @@ class WorkflowWatchHub:
         an entry with event_type DELETED means the object is gone."""
         return self._last.get((namespace, name))
 
+    def forget(self, namespace: str, name: str) -> None:
+        """Drop a workflow's cache/seq state. Called by the watch loop when
+        it finishes with a workflow: names are generateName-unique, so
+        nothing will ever wait on the key again — without this, one cache
+        entry per completed workflow accrues until the server-side TTL
+        delete (which can be 30 minutes away), a real leak at fleet rates."""
+        key = (namespace, name)
+        if key not in self._waiters:
+            self._last.pop(key, None)
+            self._seq.pop(key, None)
+
     async def start(self) -> None:
         self._sub = self.client.watch(WF_API_VERSION, WF_KIND, self.namespace)
         self._task = asyncio.ensure_future(self._consume())
@@ -79,9 +90,14 @@ class WorkflowWatchHub:
             key = (meta.get("namespace", ""), meta.get("name", ""))
             waiters = self._waiters.pop(key, ())
             self._seq[key] = self._seq.get(key, 0) + 1
-            self._last[key] = (ev["type"], ev["object"])
             if ev["type"] == "DELETED":
+                # tombstone only — holding the full object here costs ~6 KB ×
+                # (deletion rate × prune TTL) of resident memory at fleet
+                # rates, and cached() readers only look at the event type
+                self._last[key] = ("DELETED", None)
                 self._deleted.append((time.monotonic(), key))
+            else:
+                self._last[key] = (ev["type"], ev["object"])
             for fut in waiters:  # wake everyone watching it
                 if not fut.done():
                     fut.set_result(ev["type"])

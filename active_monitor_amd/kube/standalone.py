@@ -62,6 +62,9 @@ def parse_args(argv=None):
                    help="scripted-bench: fraction of CRs whose checks fail")
     p.add_argument("--engine-delay", type=float, default=0.0,
                    help="scripted-bench: simulated workflow runtime (s)")
+    p.add_argument("--engine-ttl", type=float, default=60.0,
+                   help="completed-workflow TTL seconds (Argo ttlStrategy "
+                        "equivalent; bounds server memory at fleet rates)")
     return p.parse_args(argv)
 
 
@@ -80,11 +83,12 @@ async def amain(args) -> int:
             MemoryClient(server),
             policy=bench_policy(args.remedy_frac),
             delay=args.engine_delay,
+            ttl_seconds=args.engine_ttl,
         )
     elif args.engine == "local":
         from ..workflow import LocalWorkflowEngine
 
-        engine = LocalWorkflowEngine(MemoryClient(server))
+        engine = LocalWorkflowEngine(MemoryClient(server), ttl_seconds=args.engine_ttl)
     if engine is not None:
         await engine.start()
 

@@ -48,6 +48,7 @@ async def main_async(args) -> dict:
         apiserver_proc = await asyncio.create_subprocess_exec(
             sys.executable, "-m", "active_monitor_amd.kube.standalone",
             "--engine", "scripted-bench", "--remedy-frac", "0",
+            "--engine-ttl", str(args.ttl),
             stdout=asyncio.subprocess.PIPE, stderr=asyncio.subprocess.DEVNULL,
         )
         line = await asyncio.wait_for(apiserver_proc.stdout.readline(), 60)
@@ -106,6 +107,12 @@ async def main_async(args) -> dict:
             "active_watches": rec.active_watches(),
             "queue": len(manager.queue),
             "objects": len(server) if server is not None else None,
+            # leak canaries: bounded-structure sizes
+            "hub_last": len(getattr(manager.wf_hub, "_last", ()) or ()),
+            "hub_seq": len(getattr(manager.wf_hub, "_seq", ()) or ()),
+            "agg": len(getattr(manager.recorder, "_agg", ()) or ()),
+            "timers": len(rec.repeat_timers_by_name),
+            "cache": len(manager.hc_cache),
         })
     elapsed = time.monotonic() - t0
     total = rec.completed_runs - runs_start
