@@ -2,7 +2,7 @@
 # targets, adapted to the Python stack).
 PY ?= python3
 
-.PHONY: test test-gpu bench run manifests native docker-build sweep
+.PHONY: test test-gpu bench run manifests native docker-build sweep soak e2e-kind lint
 
 test:
 	$(PY) -m pytest tests/ -q -m "not gpu"
@@ -27,3 +27,12 @@ manifests:
 
 docker-build:
 	docker build -t active-monitor-amd:latest .
+
+soak:
+	$(PY) benchmarks/soak.py --crs 1000 --repeat 5 --duration 300
+
+e2e-kind:
+	hack/e2e-kind.sh
+
+lint:
+	ruff check --select E9,F63,F7,F82 .
