@@ -13,7 +13,9 @@ import json
 import logging
 import ssl
 import time
-from typing import Any, AsyncIterator, Dict, List, Optional
+from collections import deque
+from typing import Any, AsyncIterator, Dict, List, Optional, Tuple
+from urllib.parse import urlencode, urlsplit
 
 import aiohttp
 
@@ -223,6 +225,117 @@ class _TokenBucket:
             self.last = time.monotonic()
 
 
+class _ConnPool:
+    """Persistent plain-HTTP/1.1 connections for the unary hot path.
+
+    aiohttp's client spends ~100µs of framework machinery per request
+    (middlewares, tracing, CIMultiDict headers, timer contexts); at fleet
+    rates the controller loop is request-bound, so the ~5 unary calls per
+    reconcile cycle go over raw asyncio streams with keep-alive instead.
+    Watches (streaming) and HTTPS targets stay on aiohttp."""
+
+    def __init__(self, host: str, port: int, max_conns: int = 32):
+        self.host = host
+        self.port = port
+        self.max_conns = max_conns
+        self._free: List[Tuple[asyncio.StreamReader, asyncio.StreamWriter]] = []
+        self._count = 0
+        self._waiters: "deque[asyncio.Future]" = deque()
+        self._closed = False
+
+    async def _acquire(self):
+        while True:
+            while self._free:
+                reader, writer = self._free.pop()
+                if writer.is_closing():
+                    self._count -= 1
+                    continue
+                return reader, writer
+            if self._count < self.max_conns:
+                self._count += 1
+                try:
+                    return await asyncio.open_connection(self.host, self.port)
+                except Exception:
+                    self._count -= 1
+                    raise
+            fut = asyncio.get_running_loop().create_future()
+            self._waiters.append(fut)
+            await fut
+
+    def _release(self, conn, reusable: bool) -> None:
+        if reusable and not self._closed and not conn[1].is_closing():
+            self._free.append(conn)
+        else:
+            self._count -= 1
+            try:
+                conn[1].close()
+            except Exception:
+                pass
+        while self._waiters:
+            fut = self._waiters.popleft()
+            if not fut.done():
+                fut.set_result(None)
+                break
+
+    async def request(self, method: str, target: str, headers: str,
+                      body: Optional[bytes]) -> Tuple[int, bytes]:
+        conn = await self._acquire()
+        reader, writer = conn
+        ok = False
+        try:
+            head = (
+                f"{method} {target} HTTP/1.1\r\nHost: {self.host}:{self.port}\r\n"
+                f"{headers}Content-Length: {len(body) if body else 0}\r\n\r\n"
+            ).encode("latin-1")
+            writer.write(head + body if body else head)
+            await writer.drain()
+
+            status_line = await reader.readline()
+            if not status_line:
+                raise ConnectionResetError("server closed connection")
+            status = int(status_line.split(b" ", 2)[1])
+            content_length = 0
+            chunked = False
+            keep_alive = True
+            while True:
+                line = await reader.readline()
+                if line in (b"\r\n", b"\n", b""):
+                    break
+                lower = line.lower()
+                if lower.startswith(b"content-length:"):
+                    content_length = int(line.split(b":", 1)[1])
+                elif lower.startswith(b"transfer-encoding:") and b"chunked" in lower:
+                    chunked = True
+                elif lower.startswith(b"connection:") and b"close" in lower:
+                    keep_alive = False
+            if chunked:
+                chunks = []
+                while True:
+                    size_line = await reader.readline()
+                    size = int(size_line.strip().split(b";")[0], 16)
+                    if size == 0:
+                        await reader.readline()  # trailing CRLF
+                        break
+                    chunks.append(await reader.readexactly(size))
+                    await reader.readexactly(2)  # CRLF after each chunk
+                payload = b"".join(chunks)
+            else:
+                payload = await reader.readexactly(content_length) if content_length else b""
+            ok = keep_alive
+            return status, payload
+        finally:
+            self._release(conn, ok)
+
+    def close(self) -> None:
+        self._closed = True
+        for _, writer in self._free:
+            try:
+                writer.close()
+            except Exception:
+                pass
+        self._free.clear()
+
+
 class HttpClient:
     def __init__(
         self,
@@ -242,6 +355,9 @@ class HttpClient:
         #: optional (cert_path, key_path) for client-certificate auth
         self.client_cert: tuple = (None, None)
         self._session: Optional[aiohttp.ClientSession] = None
+        # plain-HTTP unary fast path (see _ConnPool); None for https targets
+        self._pool: Optional[_ConnPool] = None
+        self._pool_headers = ""
         # qps<=0 disables throttling (benchmarks measuring the raw wire)
         self._limiter = _TokenBucket(qps, burst)
         #: unary requests issued (observability/bench; watch streams excluded)
@@ -267,8 +383,16 @@ class HttpClient:
         self._session = aiohttp.ClientSession(
             headers=headers, connector=aiohttp.TCPConnector(ssl=ssl_ctx)
         )
+        if self.base_url.startswith("http://"):
+            parts = urlsplit(self.base_url)
+            self._pool = _ConnPool(parts.hostname or "127.0.0.1", parts.port or 80)
+            self._pool_headers = "Content-Type: application/json\r\n" + (
+                f"Authorization: Bearer {self.token}\r\n" if self.token else ""
+            )
 
     async def close(self) -> None:
+        if self._pool is not None:
+            self._pool.close()
         if self._session is not None:
             await self._session.close()
 
@@ -298,6 +422,22 @@ class HttpClient:
             raise RuntimeError("HttpClient.start() must be called before requests")
         await self._limiter.acquire()
         self.request_count += 1
+        if self._pool is not None:
+            target = path + ("?" + urlencode(params) if params else "")
+            payload = json.dumps(body, separators=(",", ":")).encode() if body is not None else None
+            try:
+                status, text_b = await self._pool.request(
+                    method, target, self._pool_headers, payload
+                )
+            except (ConnectionError, asyncio.IncompleteReadError, OSError):
+                # stale keep-alive connection: one clean retry on a fresh one
+                status, text_b = await self._pool.request(
+                    method, target, self._pool_headers, payload
+                )
+            text = text_b.decode("utf-8", "replace")
+            if status >= 400:
+                raise _error_for(status, text)
+            return json.loads(text) if text else {}
         async with self._session.request(
             method, self.base_url + path,
             json=body if body is not None else None, params=params,

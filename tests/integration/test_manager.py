@@ -130,7 +130,7 @@ def test_mini_fleet_soak(run):
             async def settled():
                 return rec.completed_runs >= 150
 
-            await env.wait_for(settled, timeout=30, msg="first wave")
+            await env.wait_for(settled, timeout=45, msg="first wave")
             for i in range(150):
                 env.manager.queue.add_nowait(("health", f"fleet-{i:03d}"), {"timer"})
 
@@ -141,7 +141,7 @@ def test_mini_fleet_soak(run):
                     for o in objs
                 )
 
-            await env.wait_for(all_advanced, timeout=30, interval=0.2,
+            await env.wait_for(all_advanced, timeout=45, interval=0.2,
                                msg="second wave on every CR")
 
     run(go(), timeout=90)
