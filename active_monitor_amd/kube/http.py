@@ -490,6 +490,24 @@ class HttpClient:
         ) + "/status"
         return await self._request("PUT", path, obj)
 
+    async def patch(self, api_version: str, kind: str, namespace: str, name: str,
+                    patch: Obj, subresource: str = "") -> Obj:
+        """JSON merge-patch (kubectl patch semantics). Not on the reconcile
+        hot path — goes through the full aiohttp session for header control."""
+        path = self._object_path(api_version, kind, namespace, name)
+        if subresource:
+            path += f"/{subresource}"
+        await self._limiter.acquire()
+        self.request_count += 1
+        async with self._session.patch(
+            self.base_url + path, data=json.dumps(patch).encode(),
+            headers={"Content-Type": "application/merge-patch+json"},
+        ) as resp:
+            text = await resp.text()
+            if resp.status >= 400:
+                raise _error_for(resp.status, text)
+            return json.loads(text) if text else {}
+
     async def delete(self, api_version: str, kind: str, namespace: str, name: str) -> None:
         await self._request("DELETE", self._object_path(api_version, kind, namespace, name))
 

@@ -92,12 +92,16 @@ class ApiServerFrontend:
                 except ValueError:
                     return
                 content_length = 0
+                content_type = ""
                 while True:
                     line = await reader.readline()
                     if line in (b"\r\n", b"\n", b""):
                         break
-                    if line.lower().startswith(b"content-length:"):
+                    lower = line.lower()
+                    if lower.startswith(b"content-length:"):
                         content_length = int(line.split(b":", 1)[1])
+                    elif lower.startswith(b"content-type:"):
+                        content_type = line.split(b":", 1)[1].strip().decode("latin-1")
                 body = await reader.readexactly(content_length) if content_length else b""
 
                 parts = urlsplit(target)
@@ -107,7 +111,8 @@ class ApiServerFrontend:
                 if query.get("watch") in ("true", "1"):
                     await self._serve_watch(writer, path, query)
                     return  # watch streams own the connection
-                status, obj = self._serve_unary(method, path, query, body)
+                status, obj = self._serve_unary(method, path, query, body,
+                                                content_type)
                 payload = json.dumps(obj).encode() if obj is not None else b""
                 writer.write(
                     (
@@ -142,7 +147,7 @@ class ApiServerFrontend:
         return None
 
     def _serve_unary(self, method: str, path: str, query: dict,
-                     body: bytes) -> Tuple[int, Optional[dict]]:
+                     body: bytes, content_type: str = "") -> Tuple[int, Optional[dict]]:
         # discovery endpoints (enough for kubectl --server=<url>)
         if path in ("/api", "/api/"):
             return 200, {"kind": "APIVersions", "versions": ["v1"]}
@@ -161,7 +166,7 @@ class ApiServerFrontend:
         api_version, tail = split
         if not tail:  # /api/v1 or /apis/g/v → resource discovery
             return 200, self._resource_list(api_version)
-        return self._dispatch(method, api_version, tail, query, body)
+        return self._dispatch(method, api_version, tail, query, body, content_type)
 
     def _discovery_apis(self) -> dict:
         groups_map: dict = {}
@@ -204,7 +209,7 @@ class ApiServerFrontend:
                 "groupVersion": api_version, "resources": resources}
 
     def _dispatch(self, method: str, api_version: str, tail: str, query: dict,
-                  body: bytes) -> Tuple[int, Optional[dict]]:
+                  body: bytes, content_type: str = "") -> Tuple[int, Optional[dict]]:
         # tail forms:
         #   {plural}[/{name}[/status]]
         #   namespaces/{ns}/{plural}[/{name}[/status]]
@@ -251,6 +256,31 @@ class ApiServerFrontend:
             if method == "DELETE":
                 self.server.delete(api_version, kind, namespace, name)
                 return 200, {"kind": "Status", "status": "Success"}
+            if method == "PATCH":
+                ct = content_type.split(";")[0].strip().lower()
+                if ct == "application/json-patch+json":
+                    # RFC 6902 op lists are not supported; kubectl defaults
+                    # to strategic/merge for CRDs anyway
+                    return 415, _status_body(
+                        InvalidError("json-patch is not supported; use merge-patch")
+                    )
+                if ct == "application/apply-patch+yaml":
+                    import yaml as _yaml
+
+                    try:
+                        patch_obj = _yaml.safe_load(body)
+                    except _yaml.YAMLError:
+                        raise InvalidError("request body is not valid JSON")
+                    if not isinstance(patch_obj, dict):
+                        raise InvalidError("request body is not valid JSON")
+                    upsert = True  # server-side apply creates when absent
+                else:  # merge-patch / strategic-merge-patch
+                    patch_obj = self._parse_body(body)
+                    upsert = False
+                return 200, self.server.patch(
+                    api_version, kind, namespace, name, patch_obj,
+                    subresource=subresource, upsert=upsert,
+                )
         except InvalidError as e:
             code = 400 if e.message == "request body is not valid JSON" else e.code
             return code, _status_body(e)

@@ -203,9 +203,17 @@ def test_http_error_paths(run):
                 ) as r:
                     assert r.status == 400
                     assert (await r.json())["kind"] == "Status"
+                # PATCH with an empty body is a 400; a valid merge patch on
+                # a missing object is a 404 (PATCH itself is supported now)
                 async with s.patch(
                     base + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/health/healthchecks/x"
                 ) as r:
-                    assert r.status in (404, 405)
+                    assert r.status == 400
+                async with s.patch(
+                    base + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/health/healthchecks/x",
+                    data=b'{"spec":{"repeatAfterSec":9}}',
+                    headers={"Content-Type": "application/merge-patch+json"},
+                ) as r:
+                    assert r.status == 404
 
     run(go(), timeout=30)
