@@ -334,6 +334,10 @@ class _ConnPool:
             except Exception:
                 pass
         self._free.clear()
+        while self._waiters:  # fail queued acquirers instead of hanging them
+            fut = self._waiters.popleft()
+            if not fut.done():
+                fut.set_exception(ConnectionError("connection pool closed"))
 
 
 class HttpClient:
