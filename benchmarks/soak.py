@@ -47,7 +47,7 @@ async def main_async(args) -> dict:
 
         apiserver_proc = await asyncio.create_subprocess_exec(
             sys.executable, "-m", "active_monitor_amd.kube.standalone",
-            "--engine", "scripted-bench", "--remedy-frac", "0",
+            "--engine", "scripted-bench", "--remedy-frac", str(args.remedy_frac),
             "--engine-ttl", str(args.ttl),
             stdout=asyncio.subprocess.PIPE, stderr=asyncio.subprocess.DEVNULL,
         )
@@ -57,9 +57,11 @@ async def main_async(args) -> dict:
         client = HttpClient(url, qps=0)
         await client.start()
     else:
+        from active_monitor_amd.kube.standalone import bench_policy
+
         server = MemoryApiServer()
         client = MemoryClient(server)
-        engine = ScriptedWorkflowEngine(client, policy=lambda wf: ("Succeeded", ""),
+        engine = ScriptedWorkflowEngine(client, policy=bench_policy(args.remedy_frac),
                                         ttl_seconds=args.ttl)
         await engine.start()
     manager = Manager(client, max_workers=args.workers)
@@ -71,23 +73,37 @@ async def main_async(args) -> dict:
         "        command: [echo, ok]\n"
     )
     for i in range(args.crs):
+        # names follow the bench convention (hc-NNNNN) so the scripted
+        # engine's policy can fail the remedy-carrying fraction by name
+        is_remedy = (i % 100) < args.remedy_frac * 100
+        spec = {
+            "repeatAfterSec": args.repeat,
+            "level": "cluster",
+            "workflow": {
+                "generateName": f"hc-{i:05d}-wf-",
+                "workflowtimeout": max(args.repeat, 5),
+                "resource": {
+                    "namespace": "health",
+                    "serviceAccount": f"soak-sa-{i % 8}",
+                    "source": {"inline": inline},
+                },
+            },
+        }
+        if is_remedy:
+            spec["remedyworkflow"] = {
+                "generateName": f"hc-{i:05d}-remedy-wf-",
+                "workflowtimeout": max(args.repeat, 5),
+                "resource": {
+                    "namespace": "health",
+                    "serviceAccount": f"soak-remedy-sa-{i}",
+                    "source": {"inline": inline},
+                },
+            }
         await client.create({
             "apiVersion": "activemonitor.keikoproj.io/v1alpha1",
             "kind": "HealthCheck",
-            "metadata": {"name": f"soak-{i:05d}", "namespace": "health"},
-            "spec": {
-                "repeatAfterSec": args.repeat,
-                "level": "cluster",
-                "workflow": {
-                    "generateName": f"soak-{i:05d}-wf-",
-                    "workflowtimeout": max(args.repeat, 5),
-                    "resource": {
-                        "namespace": "health",
-                        "serviceAccount": f"soak-sa-{i % 8}",
-                        "source": {"inline": inline},
-                    },
-                },
-            },
+            "metadata": {"name": f"hc-{i:05d}", "namespace": "health"},
+            "spec": spec,
         })
 
     rec = manager.reconciler
@@ -117,8 +133,16 @@ async def main_async(args) -> dict:
     elapsed = time.monotonic() - t0
     total = rec.completed_runs - runs_start
     lat = manager.drain_latencies()
+    remedy_runs = None
+    if args.remedy_frac > 0:
+        objs = await client.list(
+            "activemonitor.keikoproj.io/v1alpha1", "HealthCheck", "health")
+        remedy_runs = sum(
+            (o.get("status") or {}).get("remedyTotalRuns", 0) for o in objs)
     result = {
         "crs": args.crs,
+        "remedy_frac": args.remedy_frac,
+        "remedy_total_runs": remedy_runs,
         "repeat_after_sec": args.repeat,
         "duration_s": round(elapsed, 1),
         "cycles": total,
@@ -156,6 +180,9 @@ def main() -> int:
     ap.add_argument("--duration", type=float, default=120.0)
     ap.add_argument("--ttl", type=float, default=30.0,
                     help="completed-workflow TTL (Argo ttlStrategy equivalent)")
+    ap.add_argument("--remedy-frac", type=float, default=0.0,
+                    help="fraction of CRs whose checks fail and carry a "
+                         "remedy workflow (steady-state remedy machinery)")
     ap.add_argument("--apiserver", choices=["wire", "memory"], default="wire",
                     help="wire: separate apiserver process over 127.0.0.1 "
                          "HTTP (the bench.py headline regime); memory: "
