@@ -92,7 +92,8 @@ def build_metrics_security(
 
 
 async def _handle(reader: asyncio.StreamReader, writer: asyncio.StreamWriter,
-                  manager, auth_token: Optional[str] = None) -> None:
+                  manager, auth_token: Optional[str] = None,
+                  serve_observability: bool = True) -> None:
     try:
         request_line = await asyncio.wait_for(reader.readline(), 10)
         if not request_line:
@@ -110,6 +111,18 @@ async def _handle(reader: asyncio.StreamReader, writer: asyncio.StreamWriter,
                 if value.lower().startswith("bearer "):
                     bearer = value[7:]
 
+        if not serve_observability and path.startswith(_PROTECTED):
+            # probe-only server: /metrics and /statusz live on the (secured)
+            # metrics endpoint, not the open probe port
+            body = b"not found"
+            writer.write(
+                (
+                    "HTTP/1.1 404 Not Found\r\nContent-Type: text/plain\r\n"
+                    f"Content-Length: {len(body)}\r\nConnection: close\r\n\r\n"
+                ).encode("latin-1") + body
+            )
+            await writer.drain()
+            return
         if auth_token is not None and path.startswith(_PROTECTED):
             if bearer != auth_token:
                 body = b"Unauthorized"
@@ -187,11 +200,17 @@ async def serve_endpoints(
     any metrics security applies to it as a whole)."""
     servers: List[asyncio.AbstractServer] = []
 
-    async def open_handler(r, w):
-        await _handle(r, w, manager)
-
     token = metrics_security.token if metrics_security else None
     ssl_ctx = metrics_security.ssl_context if metrics_security else None
+    # when a dedicated (possibly secured) metrics server exists, the probe
+    # server is probes-only: /metrics and /statusz 404 there so the authn
+    # filter cannot be sidestepped via the open port. (Equal non-ephemeral
+    # addresses mean one genuinely shared server; port 0 is always distinct.)
+    shared = health is not None and health == metrics and health[1] != 0
+    probes_only = metrics is not None and not shared
+
+    async def open_handler(r, w):
+        await _handle(r, w, manager, serve_observability=not probes_only)
 
     async def secured_handler(r, w):
         await _handle(r, w, manager, auth_token=token)

@@ -95,3 +95,69 @@ def test_token_file_is_used(run, tmp_path):
     tok.write_text("sekret\n")
     sec = build_metrics_security(True, token_file=str(tok), workdir=str(tmp_path))
     assert sec.token == "sekret"
+
+
+def test_probe_port_does_not_leak_observability(run, tmp_path):
+    """With a dedicated secured metrics server, the open probe port must not
+    serve /metrics or /statusz — the authn filter can't be sidestepped."""
+    sec = build_metrics_security(True, workdir=str(tmp_path))
+
+    async def go():
+        servers = await serve_endpoints(
+            None, health=("127.0.0.1", 0), metrics=("127.0.0.1", 0),
+            metrics_security=sec,
+        )
+        try:
+            health_port = servers[0].sockets[0].getsockname()[1]
+            assert "200 OK" in await _http_get(health_port, "/healthz")
+            assert "404" in await _http_get(health_port, "/metrics")
+            assert "404" in await _http_get(health_port, "/statusz")
+        finally:
+            for s in servers:
+                s.close()
+
+    run(go(), timeout=30)
+
+
+def test_cli_default_wires_security_into_manager(run, monkeypatch):
+    """The entrypoint's default path hands the manager a MetricsSecurity."""
+    import active_monitor_amd.engine as engine_pkg
+    from active_monitor_amd.cmd.main import run as cmd_run
+
+    captured = {}
+
+    class FakeManager:
+        def __init__(self, *a, **kw):
+            import asyncio
+
+            captured.update(kw)
+            self.fatal = asyncio.Event()
+            self.fatal_reason = None
+            self.ready = True
+
+        async def start(self):
+            pass
+
+        async def stop(self):
+            pass
+
+    monkeypatch.setattr(engine_pkg, "Manager", FakeManager)
+
+    async def go():
+        import asyncio
+
+        from active_monitor_amd.cmd.main import build_parser
+
+        args = build_parser().parse_args(
+            ["--backend", "memory", "--workflow-engine", "none",
+             "--metrics-bind-address", "127.0.0.1:0",
+             "--health-probe-bind-address", "0"]
+        )
+        stop = asyncio.Event()
+        stop.set()  # exit immediately after start
+        return await cmd_run(args, stop)
+
+    rc = run(go(), timeout=30)
+    assert rc == 0
+    sec = captured.get("metrics_security")
+    assert sec is not None and sec.token and sec.ssl_context is not None
