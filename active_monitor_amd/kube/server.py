@@ -342,6 +342,14 @@ class ApiServerFrontend:
         kind = info.kind
 
         rv_param = query.get("resourceVersion")
+        selector = {}
+        if query.get("labelSelector"):
+            from .memory import parse_label_selector
+
+            try:
+                selector = parse_label_selector(query["labelSelector"])
+            except ApiError:
+                selector = {}
         sub = self.server.watch(api_version, kind, namespace)
         self._live_subs.append(sub)
         try:
@@ -373,13 +381,19 @@ class ApiServerFrontend:
                 b"Content-Type: application/json;stream=watch\r\n"
                 b"Connection: close\r\n\r\n"
             )
+            from .memory import _labels_match
+
             for ev in replay:
                 last_rv = max(last_rv, self._ev_rv(ev))
+                if selector and not _labels_match(ev.get("object") or {}, selector):
+                    continue
                 writer.write((json.dumps(ev) + "\n").encode())
             await writer.drain()
             async for ev in sub:
                 if self._ev_rv(ev) <= last_rv:
                     continue  # already covered by the replay snapshot
+                if selector and not _labels_match(ev.get("object") or {}, selector):
+                    continue
                 writer.write((json.dumps(ev) + "\n").encode())
                 await writer.drain()
         except (ConnectionError, asyncio.CancelledError):

@@ -65,6 +65,10 @@ def parse_args():
                    help="use an external Kubernetes-REST endpoint instead of "
                         "spawning one (implies --apiserver wire; a workflow "
                         "controller must be running cluster-side)")
+    p.add_argument("--kubeconfig", default="",
+                   help="with --apiserver-url (or alone): connect via this "
+                        "kubeconfig's credentials — the real-kind-cluster "
+                        "regime BASELINE.md names")
     p.add_argument("--qps", type=float, default=0.0,
                    help="client-side rate limit (0 = unlimited, the bench "
                         "default — the reference harness likewise tunes "
@@ -165,16 +169,25 @@ async def run_rank(args, rank: int):
     apiserver_proc = None
     rtt_ms = 0.0
 
-    if args.apiserver_url or args.apiserver == "wire":
+    if args.apiserver_url or args.kubeconfig or args.apiserver == "wire":
         from active_monitor_amd.kube.http import HttpClient
 
-        if args.apiserver_url:
+        if args.kubeconfig:
+            from active_monitor_amd.kube.config import get_config
+
+            cfg = get_config(server=args.apiserver_url or "",
+                             kubeconfig=args.kubeconfig)
+            client = cfg.make_client()
+            client._limiter.qps = args.qps  # bench regime overrides
+            apiserver_desc = cfg.server
+        elif args.apiserver_url:
             url = args.apiserver_url
             apiserver_desc = url
+            client = HttpClient(url, qps=args.qps)
         else:
             apiserver_proc, url = await _spawn_apiserver(args)
             apiserver_desc = "http-subprocess-127.0.0.1"
-        client = HttpClient(url, qps=args.qps)
+            client = HttpClient(url, qps=args.qps)
         await client.start()
         rtt_ms = await _measure_rtt_ms(client)
     else:
