@@ -12,8 +12,12 @@ clientset (healthcheck_controller.go:133-137). Backends:
 from __future__ import annotations
 
 import asyncio
+import time as _time
+import zlib
 from typing import Any, Dict, List, Optional, Protocol
 
+from ..api.types import k8s_now
+from .errors import AlreadyExistsError, NotFoundError
 from .memory import MemoryApiServer, Subscription
 
 Obj = Dict[str, Any]
@@ -140,8 +144,6 @@ class EventRecorder:
             self._task = asyncio.get_running_loop().create_task(self._pump())
 
     async def _write(self, op: str, key: tuple, ev: Obj) -> None:
-        from ..kube.errors import AlreadyExistsError, NotFoundError
-
         if op == "create":
             try:
                 try:
@@ -209,11 +211,6 @@ class EventRecorder:
                 del self._agg[k]
 
     async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
-        import time as _time
-        import zlib
-
-        from ..api.types import k8s_now
-
         meta = involved.get("metadata") or {}
         ns = meta.get("namespace", "") or "default"
         obj_name = meta.get("name", "object")
