@@ -93,6 +93,7 @@ class ApiServerFrontend:
                     return
                 content_length = 0
                 content_type = ""
+                expect_continue = False
                 while True:
                     line = await reader.readline()
                     if line in (b"\r\n", b"\n", b""):
@@ -102,6 +103,11 @@ class ApiServerFrontend:
                         content_length = int(line.split(b":", 1)[1])
                     elif lower.startswith(b"content-type:"):
                         content_type = line.split(b":", 1)[1].strip().decode("latin-1")
+                    elif lower.startswith(b"expect:") and b"100-continue" in lower:
+                        expect_continue = True  # curl sends this for big bodies
+                if expect_continue:
+                    writer.write(b"HTTP/1.1 100 Continue\r\n\r\n")
+                    await writer.drain()
                 body = await reader.readexactly(content_length) if content_length else b""
 
                 parts = urlsplit(target)
