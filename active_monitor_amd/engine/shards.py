@@ -228,9 +228,13 @@ class ShardCoordinator:
     async def _loop(self) -> None:
         while True:
             await asyncio.sleep(self.renew_interval)
-            await self._scan_once()
+            if not await self._scan_once():
+                return  # home lease lost: the process is going down
 
-    async def _scan_once(self) -> None:
+    async def _scan_once(self) -> bool:
+        """One renew/adopt pass; returns False when the home lease was lost
+        (the caller stops scanning — a deposed process must not keep
+        acquiring shards while it shuts down)."""
         for shard in range(self.shard_count):
             if shard in self.owned:
                 ok = await self._renew_owned(shard)
@@ -241,7 +245,7 @@ class ShardCoordinator:
                     log.error("shard %d: HOME lease lost", shard)
                     if self.on_home_lost is not None:
                         self.on_home_lost()
-                    return  # the process is going down
+                    return False
                 log.info("shard %d: released/lost (reclaimed by owner)", shard)
                 if self.on_drop is not None:
                     await self.on_drop(shard)
@@ -252,3 +256,4 @@ class ShardCoordinator:
                                 shard, self.identity)
                     if self.on_adopt is not None:
                         await self.on_adopt(shard)
+        return True
