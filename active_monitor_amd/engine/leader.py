@@ -29,6 +29,17 @@ class LeaderElector:
         renew_interval: float = 5.0,
         retry_interval: float = 2.0,
     ):
+        import os
+
+        # ops/test override for failover latency (mirrors controller-runtime's
+        # LeaseDuration/RenewDeadline tunables the reference inherits)
+        env_lease = os.environ.get("AM_LEADER_LEASE_SECS")
+        env_renew = os.environ.get("AM_LEADER_RENEW_SECS")
+        if env_lease:
+            lease_duration = float(env_lease)
+        if env_renew:
+            renew_interval = float(env_renew)
+            retry_interval = min(retry_interval, renew_interval)
         self.client = client
         self.name = name
         self.namespace = namespace
