@@ -76,8 +76,11 @@ async def main_async(args) -> dict:
         # names follow the bench convention (hc-NNNNN) so the scripted
         # engine's policy can fail the remedy-carrying fraction by name
         is_remedy = (i % 100) < args.remedy_frac * 100
+        is_cron = (
+            not is_remedy
+            and (i % 100) < (args.remedy_frac + args.cron_frac) * 100
+        )
         spec = {
-            "repeatAfterSec": args.repeat,
             "level": "cluster",
             "workflow": {
                 "generateName": f"hc-{i:05d}-wf-",
@@ -89,6 +92,12 @@ async def main_async(args) -> dict:
                 },
             },
         }
+        if is_cron:
+            # the cron scheduler path: interval recomputed at every
+            # completion (reference :251-263 semantics)
+            spec["schedule"] = {"cron": f"@every {args.repeat}s"}
+        else:
+            spec["repeatAfterSec"] = args.repeat
         if is_remedy:
             spec["remedyworkflow"] = {
                 "generateName": f"hc-{i:05d}-remedy-wf-",
@@ -180,6 +189,9 @@ def main() -> int:
     ap.add_argument("--duration", type=float, default=120.0)
     ap.add_argument("--ttl", type=float, default=30.0,
                     help="completed-workflow TTL (Argo ttlStrategy equivalent)")
+    ap.add_argument("--cron-frac", type=float, default=0.0,
+                    help="fraction of CRs scheduled by cron (@every) instead "
+                         "of repeatAfterSec")
     ap.add_argument("--remedy-frac", type=float, default=0.0,
                     help="fraction of CRs whose checks fail and carry a "
                          "remedy workflow (steady-state remedy machinery)")
