@@ -119,11 +119,15 @@ class ApiServerFrontend:
                     return  # watch streams own the connection
                 status, obj = self._serve_unary(method, path, query, body,
                                                 content_type)
-                payload = json.dumps(obj).encode() if obj is not None else b""
+                if isinstance(obj, str):  # health probes are plain text
+                    payload, ctype = obj.encode(), "text/plain"
+                else:
+                    payload = json.dumps(obj).encode() if obj is not None else b""
+                    ctype = "application/json"
                 writer.write(
                     (
                         f"HTTP/1.1 {status} {_REASONS.get(status, 'OK')}\r\n"
-                        "Content-Type: application/json\r\n"
+                        f"Content-Type: {ctype}\r\n"
                         f"Content-Length: {len(payload)}\r\n\r\n"
                     ).encode("latin-1") + payload
                 )
@@ -166,6 +170,8 @@ class ApiServerFrontend:
                 "major": "1", "minor": "33",
                 "gitVersion": f"v1.33.0-active-monitor-amd+{__version__}",
             }
+        if path in ("/healthz", "/readyz", "/livez"):
+            return 200, "ok"  # apiserver health probes (kubectl checks these)
         split = self._split(path)
         if split is None:
             return 404, _status_body(ApiError(f"the server could not find {path}"))
@@ -259,6 +265,20 @@ class ApiServerFrontend:
                 if subresource == "status":
                     return 200, self.server.update_status(obj)
                 return 200, self.server.update(obj)
+            if method == "DELETE" and not name:
+                # deletecollection (kubectl delete --all / -l selector)
+                victims = self.server.list(
+                    api_version, kind, namespace or None, query.get("labelSelector")
+                )
+                for obj in victims:
+                    m = obj.get("metadata") or {}
+                    try:
+                        self.server.delete(api_version, kind,
+                                           m.get("namespace", ""), m.get("name", ""))
+                    except ApiError:
+                        pass  # raced another delete / cascade GC
+                return 200, {"apiVersion": api_version, "kind": kind + "List",
+                             "items": victims}
             if method == "DELETE":
                 self.server.delete(api_version, kind, namespace, name)
                 return 200, {"kind": "Status", "status": "Success"}
