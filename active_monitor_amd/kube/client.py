@@ -133,8 +133,12 @@ class EventRecorder:
         self._task: Optional[asyncio.Task] = None
         self._buffer = buffer
         self.dropped = 0
-        #: dedup/rate state per (ns, kind, name, type, reason, message)
+        #: dedup state per (ns, kind, name, type, reason, message)
         self._agg: Dict[tuple, Dict[str, Any]] = {}
+        #: spam-filter token buckets per INVOLVED OBJECT (ns, kind, name) —
+        #: client-go's EventSourceObjectSpamFilter granularity; surviving
+        #: aggregation-window rollovers keeps long-running fleets suppressed
+        self._spam: Dict[tuple, Dict[str, float]] = {}
         #: observability: wire writes suppressed by the spam filter
         self.suppressed = 0
 
@@ -209,6 +213,29 @@ class EventRecorder:
                 : len(self._agg) // 2
             ]:
                 del self._agg[k]
+        if len(self._spam) > self._AGG_MAX:
+            stale = [k for k, b in self._spam.items()
+                     if now - b["refill_at"] > self.SPAM_REFILL_SECS]
+            for k in stale:
+                del self._spam[k]
+
+    def _spend_token(self, obj_key: tuple, now: float) -> bool:
+        """Per-object write-rate token bucket (burst SPAM_BURST, refill
+        1/SPAM_REFILL_SECS — client-go EventSourceObjectSpamFilter)."""
+        bucket = self._spam.get(obj_key)
+        if bucket is None:
+            bucket = {"tokens": float(self.SPAM_BURST), "refill_at": now}
+            self._spam[obj_key] = bucket
+        elapsed = now - bucket["refill_at"]
+        bucket["refill_at"] = now
+        bucket["tokens"] = min(
+            float(self.SPAM_BURST),
+            bucket["tokens"] + elapsed / self.SPAM_REFILL_SECS,
+        )
+        if bucket["tokens"] < 1.0:
+            return False
+        bucket["tokens"] -= 1.0
+        return True
 
     async def event(self, involved: Obj, ev_type: str, reason: str, message: str) -> None:
         meta = involved.get("metadata") or {}
@@ -218,6 +245,7 @@ class EventRecorder:
         now = _time.monotonic()
         stamp = k8s_now()
 
+        obj_key = (ns, involved.get("kind"), obj_name)
         entry = self._agg.get(key)
         if entry is not None and now - entry["t0"] > self.AGG_TTL:
             entry = None  # aggregation window rolled over: start fresh
@@ -226,27 +254,19 @@ class EventRecorder:
             self._prune_agg(now)
             # deterministic per-key name so repeats update in place
             ev_name = f"{obj_name}.{zlib.crc32(repr(key).encode()):08x}"
-            entry = {
-                "t0": now, "count": 1, "first": stamp, "name": ev_name,
-                "tokens": float(self.SPAM_BURST - 1), "refill_at": now,
-            }
+            entry = {"t0": now, "count": 1, "first": stamp, "name": ev_name}
             self._agg[key] = entry
+            if not self._spend_token(obj_key, now):
+                self.suppressed += 1  # object over its write budget
+                return
             item = ("create", key,
                     self._build_event(involved, ev_type, reason, message,
                                       ev_name, ns, 1, stamp, stamp))
         else:
             entry["count"] += 1
-            # token-bucket refill (client-go spam filter shape)
-            elapsed = now - entry["refill_at"]
-            entry["refill_at"] = now
-            entry["tokens"] = min(
-                float(self.SPAM_BURST),
-                entry["tokens"] + elapsed / self.SPAM_REFILL_SECS,
-            )
-            if entry["tokens"] < 1.0:
+            if not self._spend_token(obj_key, now):
                 self.suppressed += 1  # local count keeps accruing
                 return
-            entry["tokens"] -= 1.0
             item = ("update", key,
                     self._build_event(involved, ev_type, reason, message,
                                       entry["name"], ns, entry["count"],
