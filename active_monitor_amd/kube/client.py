@@ -74,10 +74,11 @@ class MemoryClient:
     async def list(
         self, api_version: str, kind: str,
         namespace: Optional[str] = None, label_selector: Optional[str] = None,
-        snapshot_read: bool = False,
+        snapshot_read: bool = False, field_selector: Optional[str] = None,
     ) -> List[Obj]:
         await self._lat()
-        return self.server.list(api_version, kind, namespace, label_selector, snapshot_read)
+        return self.server.list(api_version, kind, namespace, label_selector,
+                                snapshot_read, field_selector)
 
     async def create(self, obj: Obj, transfer: bool = False) -> Obj:
         await self._lat()

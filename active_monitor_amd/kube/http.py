@@ -460,11 +460,13 @@ class HttpClient:
     async def list(
         self, api_version: str, kind: str,
         namespace: Optional[str] = None, label_selector: Optional[str] = None,
-        snapshot_read: bool = False,
+        snapshot_read: bool = False, field_selector: Optional[str] = None,
     ) -> List[Obj]:
         params = {}
         if label_selector:
             params["labelSelector"] = label_selector
+        if field_selector:
+            params["fieldSelector"] = field_selector
         out = await self._request(
             "GET", self._collection_path(api_version, kind, namespace), params=params
         )

@@ -78,6 +78,42 @@ def _labels_match(obj: Obj, selector: Dict[str, str]) -> bool:
     return all(labels.get(k) == v for k, v in selector.items())
 
 
+def parse_field_selector(selector: Optional[str]) -> List[Tuple[str, str, bool]]:
+    """Parse ``a.b=v,c!=w`` into (dotted-path, value, negated) terms —
+    the equality-based fieldSelector subset kubectl uses (e.g. `kubectl
+    describe` filters Events by involvedObject.name/namespace)."""
+    if not selector:
+        return []
+    out: List[Tuple[str, str, bool]] = []
+    for part in selector.split(","):
+        part = part.strip()
+        if not part:
+            continue
+        if "!=" in part:
+            k, _, v = part.partition("!=")
+            out.append((k.strip(), v.strip(), True))
+        elif "=" in part:
+            k, _, v = part.partition("=")
+            out.append((k.strip(), v.lstrip("=").strip(), False))
+        else:
+            raise InvalidError(f"unsupported field selector term: {part!r}")
+    return out
+
+
+def _fields_match(obj: Obj, terms: List[Tuple[str, str, bool]]) -> bool:
+    for path, want, negated in terms:
+        cur: Any = obj
+        for seg in path.split("."):
+            if not isinstance(cur, dict):
+                cur = None
+                break
+            cur = cur.get(seg)
+        have = "" if cur is None else str(cur)
+        if (have == want) == negated:
+            return False
+    return True
+
+
 class Subscription:
     """One watch stream. Async-iterate to receive ``{"type": ..., "object": ...}``."""
 
@@ -303,8 +339,10 @@ class MemoryApiServer:
         namespace: Optional[str] = None,
         label_selector: Optional[str] = None,
         snapshot_read: bool = False,
+        field_selector: Optional[str] = None,
     ) -> List[Obj]:
         selector = parse_label_selector(label_selector)
+        fields = parse_field_selector(field_selector)
         copier = snapshot if snapshot_read else deep_copy
         with self._lock:
             self.op_counts["list"] += 1
@@ -315,6 +353,8 @@ class MemoryApiServer:
                 if namespace is not None and ns != namespace:
                     continue
                 if selector and not _labels_match(obj, selector):
+                    continue
+                if fields and not _fields_match(obj, fields):
                     continue
                 out.append(copier(obj))
             return out

@@ -245,7 +245,8 @@ class ApiServerFrontend:
         try:
             if method == "GET" and not name:
                 items = self.server.list(
-                    api_version, kind, namespace or None, query.get("labelSelector")
+                    api_version, kind, namespace or None, query.get("labelSelector"),
+                    field_selector=query.get("fieldSelector"),
                 )
                 return 200, {
                     "apiVersion": api_version, "kind": kind + "List",

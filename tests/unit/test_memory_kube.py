@@ -206,3 +206,39 @@ def test_update_without_resource_version_is_last_write_wins():
     s.update(b)  # no conflict despite being stale
     final = s.get("activemonitor.keikoproj.io/v1alpha1", "HealthCheck", "health", "check-1")
     assert final["spec"]["repeatAfterSec"] == 20
+
+
+def test_field_selector_filtering(run):
+    """Equality fieldSelectors (the kubectl-describe Event filter shape)."""
+    from active_monitor_amd.kube.memory import MemoryApiServer
+
+    s = MemoryApiServer()
+    for i, phase in enumerate(["Succeeded", "Failed", "Succeeded"]):
+        s.create({
+            "apiVersion": "argoproj.io/v1alpha1", "kind": "Workflow",
+            "metadata": {"name": f"fs-{i}", "namespace": "health"},
+            "spec": {}, "status": {"phase": phase},
+        })
+    got = s.list("argoproj.io/v1alpha1", "Workflow", "health",
+                 field_selector="status.phase=Succeeded")
+    assert {o["metadata"]["name"] for o in got} == {"fs-0", "fs-2"}
+    got = s.list("argoproj.io/v1alpha1", "Workflow", "health",
+                 field_selector="status.phase!=Succeeded")
+    assert {o["metadata"]["name"] for o in got} == {"fs-1"}
+    got = s.list("argoproj.io/v1alpha1", "Workflow", "health",
+                 field_selector="metadata.name=fs-1,status.phase=Failed")
+    assert len(got) == 1
+
+    # the Event-by-involvedObject shape kubectl describe uses
+    s.create({
+        "apiVersion": "v1", "kind": "Event",
+        "metadata": {"name": "e1", "namespace": "health"},
+        "involvedObject": {"name": "fs-1", "kind": "Workflow"},
+        "reason": "R",
+    })
+    got = s.list("v1", "Event", "health",
+                 field_selector="involvedObject.name=fs-1")
+    assert len(got) == 1 and got[0]["metadata"]["name"] == "e1"
+    got = s.list("v1", "Event", "health",
+                 field_selector="involvedObject.name=other")
+    assert got == []
