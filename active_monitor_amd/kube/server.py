@@ -416,7 +416,27 @@ class ApiServerFrontend:
                     continue
                 writer.write((json.dumps(ev) + "\n").encode())
             await writer.drain()
-            async for ev in sub:
+            # watch budget: like a real apiserver, an expiring watch ends
+            # with a clean stream close and the client resumes from its rv
+            # (kubectl sends timeoutSeconds by default; reflectors resume)
+            deadline = None
+            if query.get("timeoutSeconds"):
+                try:
+                    deadline = (asyncio.get_running_loop().time()
+                                + float(query["timeoutSeconds"]))
+                except ValueError:
+                    pass
+            while True:
+                try:
+                    if deadline is not None:
+                        budget = deadline - asyncio.get_running_loop().time()
+                        if budget <= 0:
+                            return
+                        ev = await asyncio.wait_for(sub.__anext__(), budget)
+                    else:
+                        ev = await sub.__anext__()
+                except (StopAsyncIteration, asyncio.TimeoutError):
+                    return
                 if self._ev_rv(ev) <= last_rv:
                     continue  # already covered by the replay snapshot
                 if selector and not _labels_match(ev.get("object") or {}, selector):

@@ -162,3 +162,29 @@ def test_client_side_rate_limiter(run):
         assert loop.time() - t0 < 0.05
 
     run(go(), timeout=10)
+
+
+def test_watch_timeout_seconds_closes_stream(run):
+    """kubectl sends timeoutSeconds on watches; the server must end the
+    stream cleanly at the budget (clients resume from their rv)."""
+    import aiohttp
+
+    async def go():
+        async with Env() as env:
+            await env.client.create(make_hc(name="twatch"))
+            url = (env.frontend.url
+                   + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/"
+                     "health/healthchecks?watch=true&timeoutSeconds=0.7")
+            t0 = asyncio.get_running_loop().time()
+            lines = []
+            async with aiohttp.ClientSession() as s:
+                async with s.get(url, timeout=aiohttp.ClientTimeout(total=10)) as r:
+                    assert r.status == 200
+                    async for line in r.content:
+                        if line.strip():
+                            lines.append(line)
+            elapsed = asyncio.get_running_loop().time() - t0
+            assert lines, "initial ADDED replay missing"
+            assert elapsed < 5, f"stream did not close at the budget ({elapsed:.1f}s)"
+
+    run(go(), timeout=30)
