@@ -160,3 +160,51 @@ def test_label_selector_roundtrip(labels):
         return
     sel = ",".join(f"{k}={v}" for k, v in labels.items())
     assert parse_label_selector(sel) == labels
+
+
+# ---------------------------------------------------------------------------
+# RFC 7386 merge patch: apply(a, diff(a, b)) == b for any JSON objects
+# ---------------------------------------------------------------------------
+
+_scalars = st.one_of(st.integers(-5, 5), st.text(max_size=4), st.booleans())
+_json_obj = st.recursive(
+    st.dictionaries(st.text(min_size=1, max_size=4), _scalars, max_size=4),
+    lambda children: st.dictionaries(
+        st.text(min_size=1, max_size=4),
+        st.one_of(_scalars, children, st.lists(_scalars, max_size=3)),
+        max_size=4,
+    ),
+    max_leaves=12,
+)
+
+
+def _diff(a, b):
+    """Build the RFC 7386 merge patch transforming a into b."""
+    patch = {}
+    for k in a:
+        if k not in b:
+            patch[k] = None
+    for k, vb in b.items():
+        va = a.get(k)
+        if isinstance(va, dict) and isinstance(vb, dict):
+            sub = _diff(va, vb)
+            if sub:
+                patch[k] = sub
+        elif va != vb or k not in a:
+            patch[k] = vb
+    return patch
+
+
+@given(a=_json_obj, b=_json_obj)
+@settings(max_examples=200, deadline=None)
+def test_merge_patch_roundtrip(a, b):
+    """MemoryApiServer._merge implements RFC 7386: applying diff(a,b) to a
+    yields b (None values cannot round-trip by the RFC's own design; the
+    generator produces none)."""
+    import copy
+
+    from active_monitor_amd.kube.memory import MemoryApiServer
+
+    base = copy.deepcopy(a)
+    MemoryApiServer._merge(base, _diff(a, b))
+    assert base == b
