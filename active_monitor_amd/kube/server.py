@@ -94,6 +94,7 @@ class ApiServerFrontend:
                 content_length = 0
                 content_type = ""
                 expect_continue = False
+                want_table = False
                 while True:
                     line = await reader.readline()
                     if line in (b"\r\n", b"\n", b""):
@@ -105,6 +106,8 @@ class ApiServerFrontend:
                         content_type = line.split(b":", 1)[1].strip().decode("latin-1")
                     elif lower.startswith(b"expect:") and b"100-continue" in lower:
                         expect_continue = True  # curl sends this for big bodies
+                    elif lower.startswith(b"accept:") and b"as=table" in lower:
+                        want_table = True
                 if expect_continue:
                     writer.write(b"HTTP/1.1 100 Continue\r\n\r\n")
                     await writer.drain()
@@ -119,6 +122,9 @@ class ApiServerFrontend:
                     return  # watch streams own the connection
                 status, obj = self._serve_unary(method, path, query, body,
                                                 content_type)
+                if (want_table and method == "GET" and status == 200
+                        and isinstance(obj, dict)):
+                    obj = self._to_table(obj)
                 if isinstance(obj, str):  # health probes are plain text
                     payload, ctype = obj.encode(), "text/plain"
                 else:
@@ -324,6 +330,69 @@ class ApiServerFrontend:
             return obj
         except (ValueError, TypeError):
             raise InvalidError("request body is not valid JSON")
+
+    # -- Table responses (kubectl get) --------------------------------------
+
+    #: printcolumns per kind, mirroring the CRD's additionalPrinterColumns
+    #: (api/crd.py; reference healthcheck_types.go:71-76 printcolumn markers)
+    _PRINTCOLUMNS = {
+        "HealthCheck": [
+            ("LATEST STATUS", "string", ("status", "status")),
+            ("SUCCESS CNT  ", "string", ("status", "successCount")),
+            ("FAIL CNT", "string", ("status", "failedCount")),
+            ("REMEDY SUCCESS CNT  ", "string", ("status", "remedySuccessCount")),
+            ("REMEDY FAIL CNT", "string", ("status", "remedyFailedCount")),
+            ("Age", "date", ("metadata", "creationTimestamp")),
+        ],
+        "Workflow": [
+            ("Status", "string", ("status", "phase")),
+            ("Age", "date", ("metadata", "creationTimestamp")),
+        ],
+    }
+    _DEFAULT_COLUMNS = [("Age", "date", ("metadata", "creationTimestamp"))]
+
+    def _to_table(self, obj: dict) -> dict:
+        """meta.k8s.io/v1 Table transform — what kubectl get requests via
+        Accept: ...;as=Table. Rows carry PartialObjectMetadata objects."""
+        if obj.get("kind", "").endswith("List"):
+            items = obj.get("items", [])
+            kind = obj.get("kind", "")[:-4]
+            meta = obj.get("metadata", {})
+        elif obj.get("kind") and obj.get("metadata") is not None:
+            items = [obj]
+            kind = obj.get("kind", "")
+            meta = {}
+        else:
+            return obj  # Status bodies etc. pass through
+        cols = self._PRINTCOLUMNS.get(kind, self._DEFAULT_COLUMNS)
+        defs = [{"name": "Name", "type": "string", "format": "name"}] + [
+            {"name": n, "type": t} for (n, t, _) in cols
+        ]
+        rows = []
+        for it in items:
+            cells = [((it.get("metadata") or {}).get("name", ""))]
+            for _, _, path in cols:
+                cur = it
+                for seg in path:
+                    cur = cur.get(seg) if isinstance(cur, dict) else None
+                    if cur is None:
+                        break
+                cells.append("" if cur is None else cur)
+            rows.append({
+                "cells": cells,
+                "object": {
+                    "kind": "PartialObjectMetadata",
+                    "apiVersion": "meta.k8s.io/v1",
+                    "metadata": it.get("metadata", {}),
+                },
+            })
+        return {
+            "kind": "Table",
+            "apiVersion": "meta.k8s.io/v1",
+            "metadata": meta,
+            "columnDefinitions": defs,
+            "rows": rows,
+        }
 
     # -- watch --------------------------------------------------------------
 

@@ -217,3 +217,42 @@ def test_http_error_paths(run):
                     assert r.status == 404
 
     run(go(), timeout=30)
+
+
+def test_table_responses_for_kubectl_get(run):
+    """kubectl get requests Accept: ...;as=Table — the frontend renders the
+    CRD's printcolumns server-side (reference healthcheck_types.go:71-76)."""
+    import aiohttp
+
+    async def go():
+        async with HttpEnv() as env:
+            await env.client.create(make_hc(name="tbl-1"))
+            obj = await env.client.get(API_VERSION, "HealthCheck", "health", "tbl-1")
+            obj["status"] = {"status": "Succeeded", "successCount": 4}
+            await env.client.update_status(obj)
+            accept = ("application/json;as=Table;v=v1;g=meta.k8s.io, "
+                      "application/json")
+            async with aiohttp.ClientSession() as s:
+                base = (env.frontend.url
+                        + "/apis/activemonitor.keikoproj.io/v1alpha1/"
+                          "namespaces/health/healthchecks")
+                async with s.get(base, headers={"Accept": accept}) as r:
+                    table = await r.json()
+                assert table["kind"] == "Table"
+                names = [c["name"] for c in table["columnDefinitions"]]
+                assert names[0] == "Name" and "LATEST STATUS" in names
+                row = table["rows"][0]
+                assert row["cells"][0] == "tbl-1"
+                assert row["cells"][1] == "Succeeded"
+                assert row["cells"][2] == 4
+                assert row["object"]["kind"] == "PartialObjectMetadata"
+
+                # single-object GET renders too; plain Accept stays JSON
+                async with s.get(base + "/tbl-1", headers={"Accept": accept}) as r:
+                    one = await r.json()
+                assert one["kind"] == "Table" and len(one["rows"]) == 1
+                async with s.get(base) as r:
+                    plain = await r.json()
+                assert plain["kind"] == "HealthCheckList"
+
+    run(go(), timeout=30)
