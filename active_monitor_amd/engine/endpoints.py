@@ -155,6 +155,12 @@ async def _handle(reader: asyncio.StreamReader, writer: asyncio.StreamWriter,
                     "completed_runs": getattr(rec, "completed_runs", 0),
                     "active_watches": rec.active_watches() if rec else 0,
                     "armed_timers": len(getattr(rec, "repeat_timers_by_name", {})),
+                    "apiserver_requests": getattr(
+                        getattr(manager, "client", None), "request_count", None
+                    ),
+                    "owned_shards": sorted(
+                        getattr(getattr(manager, "coordinator", None), "owned", ())
+                    ),
                 }
             body, ctype, code = json.dumps(stats).encode(), "application/json", 200
         elif path.startswith("/healthz"):

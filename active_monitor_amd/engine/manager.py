@@ -51,6 +51,7 @@ class Manager:
         shard_ha: bool = False,
         shard_lease_duration: float = 15.0,
         shard_renew_interval: float = 5.0,
+        resync_period: float = 10 * 3600.0,
     ):
         self.client = client
         self.max_workers = max_workers
@@ -109,6 +110,10 @@ class Manager:
         self.hc_cache: dict = {}
         self._cache_synced = False
         self.enable_hc_cache = True
+        # informer resync (controller-runtime SyncPeriod, default 10h): a
+        # periodic full re-list that re-enqueues every owned key, healing any
+        # state a lost watch event could have left stale; 0 disables
+        self.resync_period = resync_period
 
     # -- lifecycle ----------------------------------------------------------
 
@@ -163,6 +168,8 @@ class Manager:
             self.reconciler.wf_hub = self.wf_hub
 
         self._tasks.append(asyncio.ensure_future(self._informer()))
+        if self.resync_period > 0:
+            self._tasks.append(asyncio.ensure_future(self._resync_loop()))
         for i in range(self.max_workers):
             self._tasks.append(asyncio.ensure_future(self._worker(i)))
         self._started.set()
@@ -296,6 +303,34 @@ class Manager:
                 else:
                     self.hc_cache[key] = obj
                 await self.queue.add(key)
+
+    async def _resync_loop(self) -> None:
+        """Periodic full re-list + re-enqueue of owned keys (the informer
+        resync controller-runtime performs every SyncPeriod). Level-triggered
+        reconciles make this a no-op when nothing diverged."""
+        while True:
+            await asyncio.sleep(self.resync_period)
+            try:
+                objs = await self.client.list(API_VERSION, HC_KIND, self.namespace)
+            except asyncio.CancelledError:
+                raise
+            except Exception as e:
+                log.warning("resync list failed: %s", e)
+                continue
+            live = set()
+            for obj in objs:
+                meta = obj.get("metadata") or {}
+                name = meta.get("name", "")
+                if not self._owns(name):
+                    continue
+                key = (meta.get("namespace", ""), name)
+                live.add(key)
+                self.hc_cache[key] = obj
+                await self.queue.add(key)
+            # heal cache entries whose DELETED event was lost
+            for key in [k for k in self.hc_cache if k not in live]:
+                self.hc_cache.pop(key, None)
+                await self.queue.add(key)  # reconcile observes the NotFound
 
     # -- workers ------------------------------------------------------------
 

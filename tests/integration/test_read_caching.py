@@ -79,3 +79,54 @@ def test_cache_serves_deletion_as_notfound(run):
         return rec.get_timer_by_name("cachedel", "health") is None
 
     run(go(), timeout=40)
+
+
+def test_periodic_resync_heals_lost_events(run):
+    """The informer resync (controller-runtime SyncPeriod) must repair state
+    a lost watch event left stale: a cache entry for a CR deleted while the
+    event was lost gets cleaned up (timer stopped) at the next resync."""
+    import asyncio
+
+    from active_monitor_amd import API_VERSION
+    from active_monitor_amd.engine import Manager
+    from active_monitor_amd.kube import MemoryApiServer, MemoryClient
+    from active_monitor_amd.workflow import ScriptedWorkflowEngine, always_succeed
+
+    async def go():
+        server = MemoryApiServer()
+        client = MemoryClient(server)
+        engine = ScriptedWorkflowEngine(client, policy=always_succeed)
+        await engine.start()
+        manager = Manager(client, max_workers=2, resync_period=0.8)
+        await manager.start()
+        rec = manager.reconciler
+        try:
+            from .conftest import make_hc
+
+            await client.create(make_hc(name="rsync-1", repeat=3600, timeout=2))
+            deadline = asyncio.get_running_loop().time() + 20
+            while rec.completed_runs < 1:
+                assert asyncio.get_running_loop().time() < deadline
+                await asyncio.sleep(0.05)
+            assert rec.get_timer_by_name("rsync-1", "health") is not None
+
+            # simulate a lost DELETED event: remove from the store without
+            # publishing (direct dict surgery — the watch never hears it)
+            key = (API_VERSION, "HealthCheck", "health", "rsync-1")
+            with server._lock:
+                server._objects.pop(key)
+
+            # the resync must discover the orphaned cache entry, reconcile
+            # the NotFound, and stop the timer
+            deadline = asyncio.get_running_loop().time() + 20
+            while rec.get_timer_by_name("rsync-1", "health") is not None:
+                assert asyncio.get_running_loop().time() < deadline, (
+                    "resync did not heal the lost delete"
+                )
+                await asyncio.sleep(0.1)
+            assert ("health", "rsync-1") not in manager.hc_cache
+        finally:
+            await manager.stop()
+            await engine.stop()
+
+    run(go(), timeout=60)
