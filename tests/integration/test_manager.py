@@ -141,8 +141,23 @@ def test_mini_fleet_soak(run):
                     for o in objs
                 )
 
-            await env.wait_for(all_advanced, timeout=45, interval=0.2,
-                               msg="second wave on every CR")
+            try:
+                await env.wait_for(all_advanced, timeout=45, interval=0.2,
+                                   msg="second wave on every CR")
+            except AssertionError:
+                objs = await env.client.list(API_VERSION, "HealthCheck", "health")
+                lagging = {
+                    o["metadata"]["name"]: (o.get("status") or {}).get(
+                        "totalHealthCheckRuns", 0)
+                    for o in objs
+                    if (o.get("status") or {}).get("totalHealthCheckRuns", 0) < 2
+                }
+                raise AssertionError(
+                    f"second wave incomplete; lagging CRs: {lagging}; "
+                    f"queue={len(env.manager.queue)} "
+                    f"watches={rec.active_watches()} "
+                    f"completed={rec.completed_runs}"
+                )
 
     run(go(), timeout=90)
 
