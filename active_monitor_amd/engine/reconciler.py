@@ -111,6 +111,13 @@ class HealthCheckReconciler:
         # informer cache, not the wire). Returns a dict, None (synced and
         # absent ⇒ NotFound) or CACHE_MISS (not synced / not covered).
         self.hc_lookup = None
+        # read-your-writes floor per key: the rv our last status write
+        # produced. A cache entry older than this is our own write's event
+        # still in flight — reading it would lose the update (a stale
+        # success/failed count rewritten over the real one). Cache reads
+        # below the floor fall through to a direct GET. This closes a
+        # lost-update race controller-runtime's cached client actually has.
+        self._written_rv: Dict[Tuple[str, str], int] = {}
         self.repeat_timers_by_name: Dict[str, RepeatTimer] = {}
         self._watch_tasks: Dict[str, Set[asyncio.Task]] = {}
         # observability for benchmarks/tests
@@ -216,20 +223,41 @@ class HealthCheckReconciler:
     # Reconcile entry (reference :170-223)
     # ------------------------------------------------------------------
 
+    @staticmethod
+    def _rv_of(obj: Optional[Dict[str, Any]]) -> int:
+        try:
+            return int((obj.get("metadata") or {}).get("resourceVersion", 0))
+        except (TypeError, ValueError, AttributeError):
+            return -1  # opaque rv: cannot order, treat as unusable floor
+
+    def note_written(self, namespace: str, name: str, obj: Optional[Dict[str, Any]]) -> None:
+        """Record the rv a write of ours produced (read-your-writes floor)."""
+        rv = self._rv_of(obj)
+        if rv > 0:
+            key = (namespace, name)
+            if rv > self._written_rv.get(key, 0):
+                self._written_rv[key] = rv
+
     async def _get_hc(self, namespace: str, name: str) -> Dict[str, Any]:
         """Read a HealthCheck through the informer cache when one is synced,
         else over the wire. Cache hits return a shallow top-level copy so the
         caller may replace top-level keys (the status-update path does)
-        without corrupting the shared cached object."""
+        without corrupting the shared cached object. Entries older than our
+        own last write (the event is still in flight) are not served — the
+        read goes direct instead, so a timer-fired repeat can never observe
+        pre-write counters and lose the update."""
         lk = self.hc_lookup
         if lk is not None:
             hit = lk(namespace, name)
             if hit is None:
+                self._written_rv.pop((namespace, name), None)
                 raise NotFoundError(
                     f'healthchecks.activemonitor.keikoproj.io "{name}" not found'
                 )
             if hit is not CACHE_MISS:
-                return dict(hit)
+                floor = self._written_rv.get((namespace, name), 0)
+                if floor == 0 or 0 < floor <= self._rv_of(hit):
+                    return dict(hit)
         return await self.client.get(API_VERSION, HC_KIND, namespace, name,
                                      snapshot_read=True)
 
@@ -785,7 +813,9 @@ class HealthCheckReconciler:
                                               hc.name, snapshot_read=True)
             fresh["status"] = hc.status.to_dict()
             try:
-                await self.client.update_status(fresh)
+                out = await self.client.update_status(fresh)
+                # read-your-writes: later cache reads must be ≥ this version
+                self.note_written(hc.namespace, hc.name, out)
                 return
             except ConflictError as e:
                 last = e
