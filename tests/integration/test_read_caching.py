@@ -130,3 +130,44 @@ def test_periodic_resync_heals_lost_events(run):
             await engine.stop()
 
     run(go(), timeout=60)
+
+
+def test_read_your_writes_floor_rejects_stale_cache(run):
+    """Deterministic repro of the lost-update race: after a status write
+    produced rv N, a cache entry at rv<N (our own event still in flight)
+    must NOT be served — the read goes direct instead."""
+    import asyncio
+
+    from active_monitor_amd import API_VERSION
+    from active_monitor_amd.engine.reconciler import HealthCheckReconciler
+    from active_monitor_amd.kube import MemoryApiServer, MemoryClient
+    from active_monitor_amd.kube.client import FakeRecorder
+
+    from .conftest import make_hc
+
+    async def go():
+        server = MemoryApiServer()
+        client = MemoryClient(server)
+        rec = HealthCheckReconciler(client, FakeRecorder())
+        created = server.create(make_hc(name="ryw"))
+        stale = {k: v for k, v in created.items()}  # rv N
+
+        # a later write bumps the floor past the stale snapshot
+        fresh = server.get(API_VERSION, "HealthCheck", "health", "ryw")
+        fresh["status"] = {"successCount": 1}
+        written = server.update_status(fresh)  # rv N+1
+        rec.note_written("health", "ryw", written)
+
+        # lookup serves the STALE entry (the event hasn't landed yet)
+        rec.hc_lookup = lambda ns, name: stale
+        got = await rec._get_hc("health", "ryw")
+        # ...but the reconciler read the CURRENT object from the server
+        assert (got.get("status") or {}).get("successCount") == 1
+        assert got["metadata"]["resourceVersion"] == written["metadata"]["resourceVersion"]
+
+        # once the cache catches up (rv ≥ floor) it is served again
+        rec.hc_lookup = lambda ns, name: written
+        got = await rec._get_hc("health", "ryw")
+        assert got["metadata"]["resourceVersion"] == written["metadata"]["resourceVersion"]
+
+    run(go(), timeout=20)
