@@ -128,7 +128,7 @@ class ApiServerFrontend:
                 if isinstance(obj, str):  # health probes are plain text
                     payload, ctype = obj.encode(), "text/plain"
                 else:
-                    payload = json.dumps(obj).encode() if obj is not None else b""
+                    payload = json.dumps(obj, separators=(",", ":")).encode() if obj is not None else b""
                     ctype = "application/json"
                 writer.write(
                     (
@@ -428,7 +428,7 @@ class ApiServerFrontend:
             except KeyError:
                 err = ApiError(f"unknown resource {tail}")
         if err is not None:
-            payload = json.dumps(_status_body(err)).encode()
+            payload = json.dumps(_status_body(err), separators=(",", ":")).encode()
             writer.write(
                 (f"HTTP/1.1 404 Not Found\r\nContent-Type: application/json\r\n"
                  f"Content-Length: {len(payload)}\r\n\r\n").encode() + payload
@@ -455,7 +455,7 @@ class ApiServerFrontend:
                 try:
                     replay = self.server.events_since(api_version, kind, namespace, rv_param)
                 except ExpiredError as e:
-                    payload = json.dumps(_status_body(e)).encode()
+                    payload = json.dumps(_status_body(e), separators=(",", ":")).encode()
                     writer.write(
                         (f"HTTP/1.1 410 Gone\r\nContent-Type: application/json\r\n"
                          f"Content-Length: {len(payload)}\r\n\r\n").encode() + payload
@@ -483,7 +483,7 @@ class ApiServerFrontend:
                 last_rv = max(last_rv, self._ev_rv(ev))
                 if selector and not _labels_match(ev.get("object") or {}, selector):
                     continue
-                writer.write((json.dumps(ev) + "\n").encode())
+                writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
             await writer.drain()
             # watch budget: like a real apiserver, an expiring watch ends
             # with a clean stream close and the client resumes from its rv
@@ -510,7 +510,7 @@ class ApiServerFrontend:
                     continue  # already covered by the replay snapshot
                 if selector and not _labels_match(ev.get("object") or {}, selector):
                     continue
-                writer.write((json.dumps(ev) + "\n").encode())
+                writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
                 await writer.drain()
         except (ConnectionError, asyncio.CancelledError):
             pass
