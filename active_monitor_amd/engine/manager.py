@@ -327,10 +327,25 @@ class Manager:
                 live.add(key)
                 self.hc_cache[key] = obj
                 await self.queue.add(key)
-            # heal cache entries whose DELETED event was lost
+            # heal cache entries whose DELETED event was lost. A key absent
+            # from the list snapshot may simply have been created AFTER the
+            # list (its watch event already consumed) — evicting it then
+            # would orphan a live CR forever, so suspects are verified with
+            # a direct read before reaping.
+            from ..kube.errors import NotFoundError
+
             for key in [k for k in self.hc_cache if k not in live]:
-                self.hc_cache.pop(key, None)
-                await self.queue.add(key)  # reconcile observes the NotFound
+                try:
+                    obj = await self.client.get(
+                        API_VERSION, HC_KIND, key[0], key[1], snapshot_read=True
+                    )
+                except NotFoundError:
+                    self.hc_cache.pop(key, None)
+                    await self.queue.add(key)  # reconcile observes the NotFound
+                except Exception as e:
+                    log.warning("resync verify %s failed: %s", key, e)
+                else:
+                    self.hc_cache[key] = obj  # alive: refresh instead
 
     # -- workers ------------------------------------------------------------
 
