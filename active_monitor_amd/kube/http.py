@@ -132,14 +132,24 @@ class HttpSubscription:
             self._queue.put_nowait({"type": "ADDED", "object": obj})
         self.relists += 1
 
+    #: server-side watch budget (client-go reflectors use 5-10 min): bounds
+    #: how long a half-open TCP connection can silently starve the informer
+    WATCH_TIMEOUT_SECS = 300
+
     async def _stream_once(self) -> None:
-        params = {"watch": "true", "allowWatchBookmarks": "true"}
+        params = {"watch": "true", "allowWatchBookmarks": "true",
+                  "timeoutSeconds": str(self.WATCH_TIMEOUT_SECS)}
         if self._resource_version:
             params["resourceVersion"] = self._resource_version
         path = self._client._collection_path(self.api_version, self.kind, self.namespace)
         async with self._client._session.get(
             self._client.base_url + path, params=params,
-            timeout=aiohttp.ClientTimeout(total=None, sock_read=None),
+            # sock_read is the client-side safety net behind the server
+            # budget: a connection that goes dead-quiet past it is aborted
+            # and the reconnect/resume path takes over
+            timeout=aiohttp.ClientTimeout(
+                total=None, sock_read=self.WATCH_TIMEOUT_SECS + 60
+            ),
         ) as resp:
             if resp.status == 410:
                 await resp.text()
