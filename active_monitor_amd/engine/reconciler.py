@@ -30,7 +30,7 @@ import asyncio
 import logging
 import time
 from dataclasses import dataclass
-from typing import Any, Dict, Optional, Set
+from typing import Any, Dict, Optional, Set, Tuple
 
 from .. import API_VERSION
 from ..api.types import HealthCheck, k8s_now, parse_k8s_time
