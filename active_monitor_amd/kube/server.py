@@ -487,31 +487,28 @@ class ApiServerFrontend:
             await writer.drain()
             # watch budget: like a real apiserver, an expiring watch ends
             # with a clean stream close and the client resumes from its rv
-            # (kubectl sends timeoutSeconds by default; reflectors resume)
-            deadline = None
+            # (kubectl sends timeoutSeconds by default; reflectors resume).
+            # Implemented as one timer that closes the subscription — the
+            # event loop stays free of per-event wait_for wrappers.
+            budget_handle = None
             if query.get("timeoutSeconds"):
                 try:
-                    deadline = (asyncio.get_running_loop().time()
-                                + float(query["timeoutSeconds"]))
+                    budget_handle = asyncio.get_running_loop().call_later(
+                        float(query["timeoutSeconds"]), sub.close
+                    )
                 except ValueError:
                     pass
-            while True:
-                try:
-                    if deadline is not None:
-                        budget = deadline - asyncio.get_running_loop().time()
-                        if budget <= 0:
-                            return
-                        ev = await asyncio.wait_for(sub.__anext__(), budget)
-                    else:
-                        ev = await sub.__anext__()
-                except (StopAsyncIteration, asyncio.TimeoutError):
-                    return
-                if self._ev_rv(ev) <= last_rv:
-                    continue  # already covered by the replay snapshot
-                if selector and not _labels_match(ev.get("object") or {}, selector):
-                    continue
-                writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
-                await writer.drain()
+            try:
+                async for ev in sub:
+                    if self._ev_rv(ev) <= last_rv:
+                        continue  # already covered by the replay snapshot
+                    if selector and not _labels_match(ev.get("object") or {}, selector):
+                        continue
+                    writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
+                    await writer.drain()
+            finally:
+                if budget_handle is not None:
+                    budget_handle.cancel()
         except (ConnectionError, asyncio.CancelledError):
             pass
         finally:
