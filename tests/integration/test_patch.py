@@ -99,3 +99,51 @@ def test_patch_over_the_wire(run):
             await fe.stop()
 
     run(go(), timeout=30)
+
+
+def test_invalid_selectors_return_k8s_errors(run):
+    """Malformed label/field selectors surface as 422 Status bodies over the
+    wire, not connection drops."""
+    import aiohttp
+
+    from active_monitor_amd.kube.memory import MemoryApiServer
+    from active_monitor_amd.kube.server import ApiServerFrontend
+
+    async def go():
+        fe = ApiServerFrontend(MemoryApiServer())
+        await fe.start()
+        try:
+            base = (fe.url + "/apis/activemonitor.keikoproj.io/v1alpha1/"
+                             "namespaces/health/healthchecks")
+            async with aiohttp.ClientSession() as s:
+                async with s.get(base + "?fieldSelector=bogus") as r:
+                    assert r.status == 422
+                    body = await r.json()
+                    assert body["kind"] == "Status"
+                async with s.get(base + "?labelSelector=no-equals-sign") as r:
+                    assert r.status == 422
+        finally:
+            await fe.stop()
+
+    run(go(), timeout=30)
+
+
+def test_table_transform_edges(run):
+    """Table rendering: empty lists, kinds without printcolumns, Status
+    bodies pass through untouched."""
+    from active_monitor_amd.kube.memory import MemoryApiServer
+    from active_monitor_amd.kube.server import ApiServerFrontend
+
+    fe = ApiServerFrontend(MemoryApiServer())
+    empty = fe._to_table({"kind": "HealthCheckList", "items": [],
+                          "metadata": {"resourceVersion": "5"}})
+    assert empty["kind"] == "Table" and empty["rows"] == []
+
+    generic = fe._to_table({
+        "kind": "ServiceAccountList",
+        "items": [{"metadata": {"name": "sa1", "creationTimestamp": "t"}}],
+    })
+    assert generic["rows"][0]["cells"] == ["sa1", "t"]  # default columns
+
+    status = {"kind": "Status", "status": "Failure", "code": 404}
+    assert fe._to_table(status) is status  # untouched

@@ -183,3 +183,56 @@ def test_home_shard_lease_loss_is_fatal(run):
             await m0.stop()
 
     run(go(), timeout=60)
+
+
+def test_double_crash_single_survivor_adopts_all(run):
+    """3 shards, 2 crash: the lone survivor must end up owning the whole
+    keyspace and cycling every CR."""
+
+    async def go():
+        server = MemoryApiServer()
+        client = MemoryClient(server)
+        engine = ScriptedWorkflowEngine(client, policy=always_succeed)
+        await engine.start()
+        buckets = _names_by_shard(9, 3)
+        for names in buckets.values():
+            for n in names:
+                await client.create(make_hc(name=n, repeat=1, timeout=2))
+
+        managers = [_mgr(server, i, count=3) for i in range(3)]
+        for m in managers:
+            await m.start()
+        try:
+            await _wait(lambda: _all_home(managers), 20, "all homes held")
+
+            # shards 1 and 2 crash simultaneously
+            for m in managers[1:]:
+                m.coordinator.crash()
+                m.reconciler.stop_all()
+                for t in m._tasks:
+                    t.cancel()
+
+            await _wait(lambda: _owns_all(managers[0]), 30, "survivor owns 0,1,2")
+            # a CR from each dead shard keeps cycling under the survivor
+            for shard in (1, 2):
+                name = buckets[shard][0]
+                before = await _runs(client, name)
+                await _wait(
+                    lambda n=name, b=before: _advanced2(client, n, b), 30,
+                    f"shard-{shard} CR cycles under the survivor",
+                )
+        finally:
+            for m in managers:
+                await m.stop()
+            await engine.stop()
+
+    async def _all_home(managers):
+        return all(m.coordinator.owned == {i} for i, m in enumerate(managers))
+
+    async def _owns_all(m):
+        return m.coordinator.owned == {0, 1, 2}
+
+    async def _advanced2(client, name, before):
+        return await _runs(client, name) > before
+
+    run(go(), timeout=90)
