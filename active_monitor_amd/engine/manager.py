@@ -183,8 +183,6 @@ class Manager:
 
     async def stop(self) -> None:
         self._stopped = True
-        if self.coordinator is not None:
-            await self.coordinator.stop()
         if self._sub is not None:
             self._sub.close()
         if getattr(self, "wf_hub", None) is not None:
@@ -197,6 +195,11 @@ class Manager:
         for t in self._tasks:
             t.cancel()
         await asyncio.gather(*self._tasks, return_exceptions=True)
+        # release shard leases only after all reconcile work has stopped —
+        # releasing first would let an adopter start driving keys this
+        # process's timers/watches could still briefly touch
+        if self.coordinator is not None:
+            await self.coordinator.stop()
         for srv in self._servers:
             srv.close()
 
