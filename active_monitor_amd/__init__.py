@@ -30,7 +30,7 @@ bounds reconcile work ("the single biggest architectural
 decision").
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 GROUP = "activemonitor.keikoproj.io"
 VERSION = "v1alpha1"
