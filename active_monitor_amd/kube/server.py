@@ -439,6 +439,7 @@ class ApiServerFrontend:
 
         rv_param = query.get("resourceVersion")
         selector = {}
+        fields = []
         if query.get("labelSelector"):
             from .memory import parse_label_selector
 
@@ -446,6 +447,13 @@ class ApiServerFrontend:
                 selector = parse_label_selector(query["labelSelector"])
             except ApiError:
                 selector = {}
+        if query.get("fieldSelector"):
+            from .memory import parse_field_selector
+
+            try:
+                fields = parse_field_selector(query["fieldSelector"])
+            except ApiError:
+                fields = []
         sub = self.server.watch(api_version, kind, namespace)
         self._live_subs.append(sub)
         try:
@@ -477,11 +485,18 @@ class ApiServerFrontend:
                 b"Content-Type: application/json;stream=watch\r\n"
                 b"Connection: close\r\n\r\n"
             )
-            from .memory import _labels_match
+            from .memory import _fields_match, _labels_match
+
+            def _matches(obj: dict) -> bool:
+                if selector and not _labels_match(obj, selector):
+                    return False
+                if fields and not _fields_match(obj, fields):
+                    return False
+                return True
 
             for ev in replay:
                 last_rv = max(last_rv, self._ev_rv(ev))
-                if selector and not _labels_match(ev.get("object") or {}, selector):
+                if not _matches(ev.get("object") or {}):
                     continue
                 writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
             await writer.drain()
@@ -502,7 +517,7 @@ class ApiServerFrontend:
                 async for ev in sub:
                     if self._ev_rv(ev) <= last_rv:
                         continue  # already covered by the replay snapshot
-                    if selector and not _labels_match(ev.get("object") or {}, selector):
+                    if not _matches(ev.get("object") or {}):
                         continue
                     writer.write((json.dumps(ev, separators=(",", ":")) + "\n").encode())
                     await writer.drain()

@@ -188,3 +188,30 @@ def test_watch_timeout_seconds_closes_stream(run):
             assert elapsed < 5, f"stream did not close at the budget ({elapsed:.1f}s)"
 
     run(go(), timeout=30)
+
+
+def test_watch_field_selector_filters_stream(run):
+    """kubectl get hc X -w watches the collection with
+    fieldSelector=metadata.name=X — only that object's events may stream."""
+    import aiohttp
+    import json as _json
+
+    async def go():
+        async with Env() as env:
+            await env.client.create(make_hc(name="fs-a"))
+            await env.client.create(make_hc(name="fs-b"))
+            url = (env.frontend.url
+                   + "/apis/activemonitor.keikoproj.io/v1alpha1/namespaces/"
+                     "health/healthchecks?watch=true&timeoutSeconds=1.2"
+                     "&fieldSelector=metadata.name%3Dfs-a")
+            names = set()
+            async with aiohttp.ClientSession() as s:
+                async with s.get(url, timeout=aiohttp.ClientTimeout(total=10)) as r:
+                    assert r.status == 200
+                    async for line in r.content:
+                        if line.strip():
+                            ev = _json.loads(line)
+                            names.add(ev["object"]["metadata"]["name"])
+            assert names == {"fs-a"}, names
+
+    run(go(), timeout=30)
